@@ -1,0 +1,162 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark: 5-input-LUT candidate scan throughput on
+the AES (Rijndael) S-box, output bit 0 (BASELINE.json's headline metric:
+"LUT-candidates/sec + final gate count, AES S-box bit 0 at 1/2/4/8 MI355X").
+
+One step = one full exhaustive 5-LUT feasibility+decomposition scan over
+all C(POOL_GATES, 5) gate combinations of a fixed synthetic pool (the AES
+target truth table under a full mask, pool grown deterministically from
+the 8 input-bit tables — random-init analog; the reference has no public
+dataset and uses the same kind of in-search pool). Work is partitioned
+across ranks by combination index (strong scaling: total work per step is
+fixed); each rank drives one MI355X through the native gfx950 kernels and
+ranks synchronize with torch.distributed (RCCL over xGMI).
+
+Run (single GPU):   python bench.py --steps 5 --warmup 2
+Run (N GPUs):       torchrun --nproc-per-node N bench.py --gpus N ...
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+POOL_GATES = 220
+POOL_SEED = 0x5B0C5EED
+SCAN_SEED = 12345
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--pool-gates", type=int, default=POOL_GATES)
+    ap.add_argument("--allow-cpu", action="store_true",
+                    help="permit the CPU scan path (dev only; GPU runs must "
+                         "use the gfx950 kernels)")
+    args = ap.parse_args()
+
+    import torch
+    import torch.distributed as dist
+
+    from sboxgates_amd import _core, models
+    from sboxgates_amd.ops import make_engine, n_choose_k
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world > 1
+
+    use_gpu = torch.cuda.is_available()
+    if distributed:
+        backend = "nccl" if use_gpu else "gloo"
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+
+    if not use_gpu and not args.allow_cpu:
+        # The driver always runs this on an MI355X box; a silent CPU run
+        # would report a meaningless number.
+        print(json.dumps({"error": "no GPU visible; pass --allow-cpu for a "
+                                   "CPU dev run"}))
+        sys.exit(3)
+
+    gpu_mode = "force" if use_gpu else "auto"
+    engine = make_engine(lut_graph=True, seed=SCAN_SEED, gpu=gpu_mode,
+                         save_states=False)
+    if use_gpu:
+        assert engine.gpu_active, "native gfx950 kernels must be active"
+
+    # Workload: AES S-box bit 0 target, full mask, deterministic pool.
+    sbox, num_inputs = models.load("rijndael")
+    engine.set_sbox(sbox, num_inputs)
+    st = engine.initial_state()
+    st.grow_pool_random(args.pool_gates, POOL_SEED)
+    assert st.num_gates == args.pool_gates
+    target = engine.target(0)
+    mask = _core.mask_for_inputs(num_inputs)
+
+    total = n_choose_k(args.pool_gates, 5)
+    begin = total * rank // world
+    end = total * (rank + 1) // world
+
+    def one_step(step_idx):
+        found, res, evaluated = engine.scan_pool(
+            5, st, target, mask, begin, end, seed=SCAN_SEED + step_idx,
+            count_all=True)
+        return evaluated
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize()
+        if distributed:
+            dist.barrier()
+
+    # Warmup.
+    for i in range(args.warmup):
+        one_step(-1 - i)
+
+    sync()
+    t0 = time.perf_counter()
+    evaluated = 0
+    for i in range(args.steps):
+        evaluated += one_step(i)
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    sync()
+
+    # Sanity: this rank must have evaluated exactly its slice each step.
+    expected = (end - begin) * args.steps
+    assert evaluated == expected, (evaluated, expected)
+
+    # Max elapsed over ranks; sum of evaluated over ranks.
+    if distributed:
+        dev = torch.device("cuda") if use_gpu else torch.device("cpu")
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed_max = float(t.item())
+        ev = torch.tensor([evaluated], dtype=torch.int64, device=dev)
+        dist.all_reduce(ev, op=dist.ReduceOp.SUM)
+        evaluated_sum = int(ev.item())
+    else:
+        elapsed_max = elapsed
+        evaluated_sum = evaluated
+
+    if rank == 0:
+        value = evaluated_sum / elapsed_max
+        out = {
+            "metric": "5LUT-candidates/sec (AES S-box bit 0)",
+            "value": value,
+            "unit": "candidates/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed_max / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "uint64-bitslice",
+            "data": "synthetic",
+            "config": {
+                "model": "rijndael S-box, output bit 0, 5-LUT exhaustive scan",
+                "pool_gates": args.pool_gates,
+                "combinations_per_step": total,
+                "global_batch": total,
+                "seq_len": 256,
+                "parallelism": f"dp{world} (combination-space split, RCCL)",
+                "device": "gpu" if use_gpu else "cpu",
+            },
+        }
+        print(json.dumps(out))
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

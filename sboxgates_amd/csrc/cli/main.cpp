@@ -1,0 +1,248 @@
+// main.cpp — sboxgates command-line driver.
+//
+// Flag-for-flag compatible with the reference CLI (sboxgates.c:43-73,
+// 895-1174): -a -c -d -g -i -l -n -o -p -s -v plus INPUT_FILE, same
+// defaults (gates = AND|OR|XOR, iterations = 1), same conflict rules
+// (-c/-d mutually exclusive, -l/-s mutually exclusive). MI355X additions:
+// --convert-hip, --seed N, --cpu / --gpu, --output-dir DIR.
+//
+// Unlike the reference there is no MPI launcher: a single process drives
+// one GPU (or the CPU path). Multi-GPU runs go through the Python layer
+// (sboxgates_amd.parallel, torchrun one process per GPU over RCCL).
+
+#include <getopt.h>
+
+#include <climits>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+
+#include "sbg/codegen.hpp"
+#include "sbg/gpu.hpp"
+#include "sbg/options.hpp"
+#include "sbg/sboxio.hpp"
+#include "sbg/search.hpp"
+#include "sbg/xmlio.hpp"
+
+namespace {
+
+const char* kVersion = "sboxgates-mi355x 1.0";
+
+void print_help(const char* prog) {
+  std::printf(
+      "Usage: %s [OPTION...] INPUT_FILE\n"
+      "Generates graphs of Boolean gates or 3-input LUTs that realize a specified\n"
+      "S-box. Generated graphs can be converted to C/CUDA/HIP source code or to\n"
+      "Graphviz DOT format.\n\n"
+      " Graph generation\n"
+      "  -a, --available-gates=gates   Specify the set of available gates\n"
+      "                                (bitfield 0-65535).\n"
+      "  -g, --graph=graph             Load graph from file as initial state.\n"
+      "                                (For use with -o.)\n"
+      "  -i, --iterations=iterations   Set number of iterations per step.\n"
+      "  -l, --lut                     Generate LUT graph. Results in smaller\n"
+      "                                graphs but takes significantly more time.\n"
+      "  -n, --append-not              Try to generate more boolean functions by\n"
+      "                                appending NOT gates.\n"
+      "  -o, --single-output=output    Generate single-output graph for specified\n"
+      "                                output.\n"
+      "  -p, --permute=value           Permute the input S-box by XORing it with\n"
+      "                                value.\n"
+      "  -s, --sat-metric              Use SAT metric.\n"
+      "  -v, --verbose                 Increase verbosity.\n\n"
+      " Graph conversion\n"
+      "  -c, --convert-c               Convert input file to a C or CUDA function.\n"
+      "  -d, --convert-dot             Convert input file to a DOT digraph.\n"
+      "      --convert-hip             Convert input file to a HIP device function.\n\n"
+      " MI355X engine\n"
+      "      --seed=N                  Deterministic RNG seed.\n"
+      "      --cpu                     Disable the GPU path.\n"
+      "      --gpu                     Require the GPU path (fail if no device).\n"
+      "      --output-dir=DIR          Directory for XML checkpoint files.\n\n"
+      "  -?, --help                    Give this help list.\n"
+      "  -V, --version                 Print program version.\n",
+      prog);
+}
+
+int fail(const char* msg, const char* arg) {
+  if (arg != nullptr) {
+    std::fprintf(stderr, "%s: %s\n", msg, arg);
+  } else {
+    std::fprintf(stderr, "%s\n", msg);
+  }
+  return 1;
+}
+
+}  // namespace
+
+int main(int argc, char** argv) {
+  sbg::options opt;
+  opt.set_avail_gates(sbg::DEFAULT_GATE_BITFIELD);
+
+  enum { OPT_HIP = 1000, OPT_SEED, OPT_CPU, OPT_GPU, OPT_OUTDIR, OPT_HELP };
+  static const struct option long_opts[] = {
+      {"available-gates", required_argument, nullptr, 'a'},
+      {"convert-c", no_argument, nullptr, 'c'},
+      {"convert-dot", no_argument, nullptr, 'd'},
+      {"graph", required_argument, nullptr, 'g'},
+      {"iterations", required_argument, nullptr, 'i'},
+      {"lut", no_argument, nullptr, 'l'},
+      {"append-not", no_argument, nullptr, 'n'},
+      {"single-output", required_argument, nullptr, 'o'},
+      {"permute", required_argument, nullptr, 'p'},
+      {"sat-metric", no_argument, nullptr, 's'},
+      {"verbose", no_argument, nullptr, 'v'},
+      {"convert-hip", no_argument, nullptr, OPT_HIP},
+      {"seed", required_argument, nullptr, OPT_SEED},
+      {"cpu", no_argument, nullptr, OPT_CPU},
+      {"gpu", no_argument, nullptr, OPT_GPU},
+      {"output-dir", required_argument, nullptr, OPT_OUTDIR},
+      {"help", no_argument, nullptr, OPT_HELP},
+      {"version", no_argument, nullptr, 'V'},
+      {nullptr, 0, nullptr, 0}};
+
+  int ch;
+  char* endptr = nullptr;
+  long v;
+  while ((ch = getopt_long(argc, argv, "a:cdg:i:lno:p:svV", long_opts, nullptr)) != -1) {
+    switch (ch) {
+      case 'a':
+        v = std::strtol(optarg, &endptr, 10);
+        if (*endptr != '\0' || v <= 0 || v > 65535) {
+          return fail("Bad available gates value", optarg);
+        }
+        opt.set_avail_gates(static_cast<sbg::u32>(v));
+        break;
+      case 'c': opt.output_c = true; break;
+      case 'd': opt.output_dot = true; break;
+      case 'g': opt.gfname = optarg; break;
+      case 'i':
+        v = std::strtol(optarg, &endptr, 10);
+        if (*endptr != '\0' || v < 1) return fail("Bad iterations value", optarg);
+        opt.iterations = static_cast<int>(v);
+        break;
+      case 'l': opt.lut_graph = true; break;
+      case 'n': opt.try_nots = true; break;
+      case 'o':
+        v = std::strtol(optarg, &endptr, 10);
+        if (*endptr != '\0' || v < 0 || v > 7) return fail("Bad output value", optarg);
+        opt.oneoutput = static_cast<int>(v);
+        break;
+      case 'p':
+        v = std::strtol(optarg, &endptr, 10);
+        if (*endptr != '\0' || v < 0 || v > 255) {
+          return fail("Bad permutation value", optarg);
+        }
+        opt.permute = static_cast<int>(v);
+        break;
+      case 's': opt.metric = sbg::METRIC_SAT; break;
+      case 'v': opt.verbosity += 1; break;
+      case OPT_HIP: opt.output_hip = true; break;
+      case OPT_SEED:
+        opt.seeded = true;
+        opt.seed = std::strtoull(optarg, &endptr, 0);
+        if (*endptr != '\0') return fail("Bad seed value", optarg);
+        break;
+      case OPT_CPU: opt.gpu = sbg::GPU_OFF; break;
+      case OPT_GPU: opt.gpu = sbg::GPU_FORCE; break;
+      case OPT_OUTDIR: opt.output_dir = optarg; break;
+      case 'V': std::printf("%s\n", kVersion); return 0;
+      case OPT_HELP: print_help(argv[0]); return 0;
+      default: return 1;
+    }
+  }
+
+  if (optind < argc) opt.fname = argv[optind];
+
+  // Conflict rules (parity: sboxgates.c:958-968).
+  int conv = (opt.output_c ? 1 : 0) + (opt.output_dot ? 1 : 0) + (opt.output_hip ? 1 : 0);
+  if (conv > 1) return fail("Cannot combine conversion options", nullptr);
+  if (opt.lut_graph && opt.metric == sbg::METRIC_SAT) {
+    return fail("SAT metric can not be combined with LUT graph generation", nullptr);
+  }
+  if (opt.fname.empty()) return fail("Input file name argument missing", nullptr);
+
+  opt.derive_function_lists();
+
+  try {
+    // Conversion mode: input file is a graph XML (parity: sboxgates.c:1096-1113).
+    if (conv == 1) {
+      sbg::state st;
+      std::string err;
+      if (!sbg::load_state(opt.fname, &st, &err)) {
+        return fail("Error when reading state file", err.c_str());
+      }
+      if (opt.output_dot) {
+        std::fputs(sbg::graph_to_dot(st).c_str(), stdout);
+        return 0;
+      }
+      sbg::codegen_lang lang = opt.output_hip ? sbg::LANG_HIP : sbg::LANG_AUTO;
+      std::string src = sbg::graph_to_source(st, lang, &err);
+      if (src.empty()) return fail("Conversion failed", err.c_str());
+      std::fputs(src.c_str(), stdout);
+      return 0;
+    }
+
+    // Search mode.
+    sbg::u8 sbox[256];
+    sbg::u32 num_inputs = 0;
+    std::string err;
+    if (!sbg::load_sbox_file(opt.fname, opt.permute, sbox, &num_inputs, &err)) {
+      return fail("Error loading S-box", err.c_str());
+    }
+    if (opt.verbosity >= 2) {
+      std::printf("Loaded %u input S-box:\n", num_inputs);
+      for (int i = 0; i < (1 << num_inputs); i++) {
+        std::printf("%02x%s", sbox[i], (i + 1) % 16 ? " " : "\n");
+      }
+    }
+
+    sbg::Engine engine(opt);
+    engine.set_sbox(sbox, static_cast<int>(num_inputs));
+
+    if (opt.verbosity >= 1) {
+      std::printf("Available gates: NOT ");
+      for (int i = 0; opt.avail_gates[i].num_inputs != 0; i++) {
+        std::printf("%s ", sbg::gate_name[opt.avail_gates[i].fun]);
+      }
+      std::printf("\nGenerated gates: ");
+      for (int i = 0; opt.avail_not[i].num_inputs != 0; i++) {
+        std::printf("%s ", sbg::gate_name[opt.avail_not[i].fun]);
+      }
+      std::printf("\nGenerated 3-input gates: ");
+      for (int i = 0; i < opt.num_avail_3; i++) {
+        std::printf("%02x ", opt.avail_3[i].fun);
+      }
+      std::printf("\n");
+      std::printf("GPU: %s\n", engine.gpu_active() ? "active" : "off (CPU path)");
+    }
+
+    if (opt.oneoutput >= engine.num_outputs()) {
+      std::fprintf(stderr,
+                   "Error: Can't generate output bit %d. Target S-box only has %d "
+                   "outputs.\n",
+                   opt.oneoutput, engine.num_outputs());
+      return 1;
+    }
+
+    sbg::state st;
+    if (opt.gfname.empty()) {
+      engine.initial_state(st);
+    } else if (!sbg::load_state(opt.gfname, &st, &err)) {
+      return fail("Error when reading state file", err.c_str());
+    } else {
+      std::printf("Loaded %s.\n", opt.gfname.c_str());
+    }
+
+    if (opt.oneoutput != -1) {
+      engine.generate_graph_one_output(st);
+    } else {
+      engine.generate_graph(st);
+    }
+  } catch (const std::exception& e) {
+    std::fprintf(stderr, "%s\n", e.what());
+    return 1;
+  }
+  return 0;
+}

@@ -1,0 +1,851 @@
+// kernels.hip — CDNA4 (gfx950) kernels for the 3/5/7-LUT candidate scans,
+// plus the GpuEngine host wrapper.
+//
+// Design (MI355X-first; see SURVEY.md §2.3 and the repo README):
+//  * One thread evaluates one candidate combination on 256-bit truth
+//    tables held as 4 u64 words. The reference runs these scans as CPU
+//    loops across MPI ranks (lut.c:116-487); here the per-candidate
+//    feasibility test is restructured as incremental cell algebra
+//    (sbg/lutcover.hpp) with a per-u early exit.
+//  * The live gate pool (<= 500 x 32 B = 16 KB) is staged in LDS once per
+//    workgroup; the shared outer-prefix cells (8 for 5-LUT, 16 for 7-LUT)
+//    are computed once per prefix and broadcast-read from LDS, so the
+//    inner loop touches only the candidate's own 2-3 tables.
+//  * Work distribution: workgroups pull combination *prefixes* (triples
+//    for K5, quadruples for K7) from a device-scope atomic queue — the
+//    per-prefix work varies by orders of magnitude, and the dequeue
+//    primitive costs ~0.25-1 us (MI355X_MICROARch price list), negligible
+//    against per-prefix work. Grids are sized >> 256 CUs.
+//  * Early exit: winner election by atomicCAS on a device-scope lock; an
+//    agent-scope abort flag (sc1, L1-bypassing load) is polled on a coarse
+//    cadence. Any valid winner is acceptable (the reference is equally
+//    nondeterministic across ranks, lut.c:213-218).
+//  * No i64 division in device code (it is software-emulated): binomials
+//    use closed forms with constant divisors (compiled to multiply-high).
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "sbg/comb.hpp"
+#include "sbg/gpu.hpp"
+#include "sbg/lutcover.hpp"
+#include "sbg/rng.hpp"
+
+namespace sbg {
+
+#define SBG_HIP_CHECK(expr)                                                    \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      throw std::runtime_error(std::string("HIP error: ") +                    \
+                               hipGetErrorString(_e) + " at " #expr);          \
+    }                                                                          \
+  } while (0)
+
+namespace {
+
+// ---------------------------------------------------------------------------
+// Device-side control block.
+// ---------------------------------------------------------------------------
+struct DevCtl {
+  u32 abort;
+  u32 lock;
+  u32 found;
+  u32 pad;
+  u16 res[10];
+  u16 pad2[3];
+  unsigned long long evaluated;
+  unsigned long long queue;      // prefix dequeue counter
+  unsigned long long hit_count;  // K7 filter
+  u32 overflow;                  // K7 filter: hit buffer overflowed
+  u32 pad3;
+};
+
+struct Hit7 {
+  u64 p1[2];
+  u64 p0[2];
+  u16 nums[7];
+  u16 pad;
+};
+
+// Closed-form binomials; constant divisors compile to multiply-high.
+__device__ __forceinline__ i64 cf2(i64 x) { return x < 2 ? 0 : x * (x - 1) / 2; }
+__device__ __forceinline__ i64 cf3(i64 x) {
+  return x < 3 ? 0 : x * (x - 1) * (x - 2) / 6;
+}
+__device__ __forceinline__ i64 cf4(i64 x) {
+  return x < 4 ? 0 : x * (x - 1) * (x - 2) * (x - 3) / 24;
+}
+__device__ __forceinline__ i64 cf5(i64 x) {
+  return x < 5 ? 0 : x * (x - 1) * (x - 2) * (x - 3) * (x - 4) / 120;
+}
+__device__ __forceinline__ i64 cf6(i64 x) {
+  return x < 6 ? 0 : x * (x - 1) * (x - 2) * (x - 3) * (x - 4) * (x - 5) / 720;
+}
+__device__ __forceinline__ i64 cf7(i64 x) {
+  if (x < 7) return 0;
+  u64 p = static_cast<u64>(x) * (x - 1) * (x - 2) * (x - 3);
+  p *= static_cast<u64>(x - 4) * (x - 5) / 2;  // keep the product in range
+  p *= static_cast<u64>(x - 6);
+  return static_cast<i64>(p / 2520);
+}
+
+// Largest a in [0, n-k] with C(n,k) - C(n-a,k) <= r, i.e. the first element
+// of the rank-r k-combination of [0,n). Binary search over a monotone
+// closed form (~log2(n) probes).
+template <i64 (*CF)(i64)>
+__device__ __forceinline__ int first_of_rank(i64 r, int n, i64 total) {
+  int lo = 0, hi = n;  // invariant: prefix(lo) <= r < prefix(hi)
+  while (hi - lo > 1) {
+    int mid = (lo + hi) >> 1;
+    i64 prefix = total - CF(n - mid);
+    if (prefix <= r) lo = mid; else hi = mid;
+  }
+  return lo;
+}
+
+// Flat pair index -> (d, e), 0 <= d < e < m (lexicographic d-major).
+__device__ __forceinline__ void dev_decode_pair(i64 q, int m, int* d, int* e) {
+  double mm = static_cast<double>(m) - 0.5;
+  double disc = mm * mm - 2.0 * static_cast<double>(q) - 0.75;
+  int dd = static_cast<int>(mm - 0.5 - __builtin_sqrt(disc > 0.0 ? disc : 0.0));
+  if (dd < 0) dd = 0;
+  if (dd > m - 2) dd = m - 2;
+  // S(d) = d*m - d*(d+1)/2; fix up to the exact row.
+  i64 S = static_cast<i64>(dd) * m - static_cast<i64>(dd) * (dd + 1) / 2;
+  while (dd > 0 && S > q) { dd--; S -= m - 1 - dd; }
+  while (S + (m - 1 - dd) <= q) { S += m - 1 - dd; dd++; }
+  *d = dd;
+  *e = static_cast<int>(q - S) + dd + 1;
+}
+
+// Flat triple index -> (d, e, f), 0 <= d < e < f < m.
+__device__ __forceinline__ void dev_decode_triple(i64 q, int m, int* d, int* e,
+                                                  int* f) {
+  int dd = first_of_rank<cf3>(q, m, cf3(m));
+  i64 rem = q - (cf3(m) - cf3(m - dd));
+  int e2, f2;
+  dev_decode_pair(rem, m - dd - 1, &e2, &f2);
+  *d = dd;
+  *e = dd + 1 + e2;
+  *f = dd + 1 + f2;
+}
+
+__device__ __forceinline__ u64 dev_rnd(u64 seed, u64 idx) {
+  return hash_mix64(seed ^ (idx * 0x9E3779B97F4A7C15ULL));
+}
+
+__device__ __forceinline__ bool dev_abort(const DevCtl* ctl) {
+  return __hip_atomic_load(&ctl->abort, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT) != 0;
+}
+
+// Publish a winner: single writer wins the lock; the rest skip. Host reads
+// after stream sync, so plain stores suffice for the payload; the abort
+// flag is agent-scope so other workgroups see it promptly.
+__device__ __forceinline__ void dev_publish(DevCtl* ctl, const u16 res[10]) {
+  if (atomicCAS(&ctl->lock, 0u, 1u) == 0u) {
+    for (int i = 0; i < 10; i++) ctl->res[i] = res[i];
+    __threadfence();
+    ctl->found = 1;
+    __hip_atomic_store(&ctl->abort, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  }
+}
+
+struct ScanArgs {
+  const ttable* pool;  // device pool (n tables)
+  DevCtl* ctl;
+  Hit7* hits;          // K7 only
+  u64 hit_cap;         // K7 only
+  ttable T1, T0;
+  int n;
+  i64 begin, end;
+  u64 excl;  // excluded gate ids < 64
+  u64 seed;
+  int count_all;
+};
+
+constexpr int SCAN_BLOCK = 256;
+
+// ---------------------------------------------------------------------------
+// K2 — 3-LUT scan. Grid-stride over combination ranks; per-thread decode by
+// binary search on closed-form binomials; cells evaluated directly (the
+// reference's serial rank-0 loop, lut.c:501-523).
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(SCAN_BLOCK) k_scan3(ScanArgs args) {
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * 4];
+  const int n = args.n;
+  for (int i = threadIdx.x; i < n * 4; i += blockDim.x) {
+    s_pool[i] = reinterpret_cast<const u64*>(args.pool)[i];
+  }
+  __syncthreads();
+
+  DevCtl* ctl = args.ctl;
+  const i64 total3 = cf3(n);
+  u64 local_eval = 0;
+  const i64 stride = static_cast<i64>(gridDim.x) * blockDim.x;
+  i64 idx = args.begin + blockIdx.x * static_cast<i64>(blockDim.x) + threadIdx.x;
+  int tick = 0;
+
+  for (; idx < args.end; idx += stride) {
+    if (((tick++) & 255) == 0 && !args.count_all && dev_abort(ctl)) break;
+    // Decode rank -> (a, b, c).
+    int a = first_of_rank<cf3>(idx, n, total3);
+    i64 rem = idx - (total3 - cf3(n - a));
+    int b2, c2;
+    dev_decode_pair(rem, n - a - 1, &b2, &c2);
+    int b = a + 1 + b2, c = a + 1 + c2;
+
+    local_eval++;
+    const ttable* ta = reinterpret_cast<const ttable*>(&s_pool[a * 4]);
+    const ttable* tb = reinterpret_cast<const ttable*>(&s_pool[b * 4]);
+    const ttable* tc = reinterpret_cast<const ttable*>(&s_pool[c * 4]);
+    u32 p1, p0;
+    if (lut3_p_masks(*ta, *tb, *tc, args.T1, args.T0, &p1, &p0)) {
+      if (!args.count_all) {
+        u8 func = lut3_function_from_p(p1, p0, dev_rnd(args.seed, idx));
+        u16 res[10] = {};
+        res[0] = func;
+        res[1] = static_cast<u16>(a);
+        res[2] = static_cast<u16>(b);
+        res[3] = static_cast<u16>(c);
+        dev_publish(ctl, res);
+        break;
+      }
+    }
+  }
+  // Per-block evaluated reduction.
+  __shared__ unsigned long long s_eval;
+  if (threadIdx.x == 0) s_eval = 0;
+  __syncthreads();
+  atomicAdd(&s_eval, static_cast<unsigned long long>(local_eval));
+  __syncthreads();
+  if (threadIdx.x == 0) atomicAdd(&ctl->evaluated, s_eval);
+}
+
+// ---------------------------------------------------------------------------
+// K3 — 5-LUT scan. Workgroups dequeue triple prefixes; the 8 prefix cells
+// (masked with target-1/target-0) are computed cooperatively into LDS; each
+// thread then screens (d, e) pairs with a per-u early exit and runs the
+// 2-coloring decomposition solver on survivors (replacing the reference's
+// 10 x 256-function brute force, lut.c:174-246).
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(SCAN_BLOCK) k_scan5(ScanArgs args) {
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * 4];
+  __shared__ alignas(16) u64 s_H1[8][4];
+  __shared__ alignas(16) u64 s_H0[8][4];
+  __shared__ i64 s_info[4];  // base rank, pair lo, pair hi, m
+  __shared__ int s_abc[4];   // a, b, c, skip-flag
+  __shared__ unsigned long long s_eval;
+
+  const int n = args.n;
+  for (int i = threadIdx.x; i < n * 4; i += blockDim.x) {
+    s_pool[i] = reinterpret_cast<const u64*>(args.pool)[i];
+  }
+  if (threadIdx.x == 0) s_eval = 0;
+
+  DevCtl* ctl = args.ctl;
+  const i64 total3 = cf3(n);
+  // Triple range covered by [begin, end): ranks group by leading triple.
+  // Host passes the first triple's index via queue start = 0 and we add
+  // t_begin here, computed from begin.
+  u64 local_eval = 0;
+
+  for (;;) {
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      // Dequeue the next triple (device-scope).
+      unsigned long long t = __hip_atomic_fetch_add(&ctl->queue, 1ULL,
+                                                    __ATOMIC_RELAXED,
+                                                    __HIP_MEMORY_SCOPE_AGENT);
+      s_info[0] = -1;
+      if (!(!args.count_all && dev_abort(ctl))) {
+        i64 tidx = static_cast<i64>(t);
+        if (tidx < total3) {
+          int a = first_of_rank<cf3>(tidx, n, total3);
+          i64 rem = tidx - (total3 - cf3(n - a));
+          int b2, c2;
+          dev_decode_pair(rem, n - a - 1, &b2, &c2);
+          int b = a + 1 + b2, c = a + 1 + c2;
+          // Base rank of (a,b,c,c+1,c+2) in C(n,5) via hockey-stick sums.
+          i64 base = (cf5(n) - cf5(n - a)) + (cf4(n - a - 1) - cf4(n - b)) +
+                     (cf3(n - b - 1) - cf3(n - c));
+          int m = n - 1 - c;  // remaining gates c+1..n-1
+          i64 npairs = cf2(m);
+          i64 lo = args.begin > base ? args.begin - base : 0;
+          i64 hi = args.end - base < npairs ? args.end - base : npairs;
+          bool excl_prefix =
+              args.excl != 0 &&
+              (((a < 64) && ((args.excl >> a) & 1)) ||
+               ((b < 64) && ((args.excl >> b) & 1)) ||
+               ((c < 64) && ((args.excl >> c) & 1)));
+          if (base >= args.end) {
+            s_info[0] = -1;  // past the range: stop
+          } else if (lo >= hi || excl_prefix) {
+            s_info[0] = -2;  // nothing to do for this triple; next
+          } else {
+            s_info[0] = base;
+            s_info[1] = lo;
+            s_info[2] = hi;
+            s_info[3] = m;
+            s_abc[0] = a;
+            s_abc[1] = b;
+            s_abc[2] = c;
+          }
+        }
+      }
+    }
+    __syncthreads();
+    if (s_info[0] == -1) break;
+    if (s_info[0] == -2) continue;
+
+    const int a = s_abc[0], b = s_abc[1], c = s_abc[2];
+    const int m = static_cast<int>(s_info[3]);
+    const i64 base = s_info[0];
+    const i64 lo = s_info[1], hi = s_info[2];
+
+    // Cooperative prefix-cell build: 64 items = (u, word, which).
+    if (threadIdx.x < 64) {
+      int u = threadIdx.x & 7;
+      int w = (threadIdx.x >> 3) & 3;
+      bool is1 = threadIdx.x >= 32;
+      u64 ca = (u & 4) ? s_pool[a * 4 + w] : ~s_pool[a * 4 + w];
+      u64 cb = (u & 2) ? s_pool[b * 4 + w] : ~s_pool[b * 4 + w];
+      u64 cc = (u & 1) ? s_pool[c * 4 + w] : ~s_pool[c * 4 + w];
+      u64 cell = ca & cb & cc;
+      if (is1) {
+        s_H1[u][w] = cell & args.T1.w[w];
+      } else {
+        s_H0[u][w] = cell & args.T0.w[w];
+      }
+    }
+    __syncthreads();
+
+    int it = 0;
+    for (i64 q = lo + threadIdx.x; q < hi; q += blockDim.x) {
+      if (((it++) & 31) == 0 && !args.count_all && dev_abort(ctl)) break;
+      int d2, e2;
+      dev_decode_pair(q, m, &d2, &e2);
+      int d = c + 1 + d2, e = c + 1 + e2;
+      if (args.excl != 0) {
+        if ((d < 64 && ((args.excl >> d) & 1)) || (e < 64 && ((args.excl >> e) & 1))) {
+          continue;
+        }
+      }
+      local_eval++;
+
+      const u64* td = &s_pool[d * 4];
+      const u64* te = &s_pool[e * 4];
+      u64 td_[4], te_[4], ntd_[4], nte_[4];
+#pragma unroll
+      for (int w = 0; w < 4; w++) {
+        td_[w] = td[w];
+        te_[w] = te[w];
+        ntd_[w] = ~td_[w];
+        nte_[w] = ~te_[w];
+      }
+
+      u32 p1 = 0, p0 = 0;
+      bool ok = true;
+#pragma unroll
+      for (int u = 0; u < 8; u++) {
+        u64 r11_1 = 0, r10_1 = 0, r01_1 = 0, r00_1 = 0;
+        u64 r11_0 = 0, r10_0 = 0, r01_0 = 0, r00_0 = 0;
+#pragma unroll
+        for (int w = 0; w < 4; w++) {
+          const u64 h1 = s_H1[u][w];
+          const u64 h0 = s_H0[u][w];
+          const u64 a1 = h1 & td_[w];
+          const u64 na1 = h1 & ntd_[w];
+          const u64 a0 = h0 & td_[w];
+          const u64 na0 = h0 & ntd_[w];
+          r11_1 |= a1 & te_[w];
+          r10_1 |= a1 & nte_[w];
+          r01_1 |= na1 & te_[w];
+          r00_1 |= na1 & nte_[w];
+          r11_0 |= a0 & te_[w];
+          r10_0 |= a0 & nte_[w];
+          r01_0 |= na0 & te_[w];
+          r00_0 |= na0 & nte_[w];
+        }
+        if ((r11_1 && r11_0) || (r10_1 && r10_0) || (r01_1 && r01_0) ||
+            (r00_1 && r00_0)) {
+          ok = false;
+          break;
+        }
+        const u32 cbase = static_cast<u32>(u) << 2;
+        const u32 b1 = (r11_1 != 0 ? 8u : 0u) | (r10_1 != 0 ? 4u : 0u) |
+                       (r01_1 != 0 ? 2u : 0u) | (r00_1 != 0 ? 1u : 0u);
+        const u32 b0 = (r11_0 != 0 ? 8u : 0u) | (r10_0 != 0 ? 4u : 0u) |
+                       (r01_0 != 0 ? 2u : 0u) | (r00_0 != 0 ? 1u : 0u);
+        p1 |= b1 << cbase;
+        p0 |= b0 << cbase;
+      }
+      if (!ok) continue;
+
+      u8 fo, fi;
+      int split;
+      if (lut5_solve_from_p(p1, p0, dev_rnd(args.seed, base + q), &fo, &fi, &split)) {
+        if (!args.count_all) {
+          const u16 nums[5] = {static_cast<u16>(a), static_cast<u16>(b),
+                               static_cast<u16>(c), static_cast<u16>(d),
+                               static_cast<u16>(e)};
+          const u8* sp = SPLITS5[split];
+          u16 res[10] = {};
+          res[0] = fo;
+          res[1] = fi;
+          for (int j = 0; j < 3; j++) res[2 + j] = nums[sp[j]];
+          res[5] = nums[sp[3]];
+          res[6] = nums[sp[4]];
+          dev_publish(ctl, res);
+        }
+      }
+    }
+  }
+
+  atomicAdd(&s_eval, static_cast<unsigned long long>(local_eval));
+  __syncthreads();
+  if (threadIdx.x == 0) atomicAdd(&ctl->evaluated, s_eval);
+}
+
+// ---------------------------------------------------------------------------
+// K4a — 7-LUT feasibility filter. Workgroups dequeue quadruple prefixes;
+// 16 prefix cells in LDS; threads screen (e, f, g) triples; survivors'
+// p-masks land in the hit buffer (no 100k cap — 288 GB HBM keeps the whole
+// frontier resident; overflow of the per-chunk buffer is reported and the
+// host re-scans a smaller range).
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(SCAN_BLOCK) k_scan7_filter(ScanArgs args) {
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * 4];
+  __shared__ alignas(16) u64 s_H1[16][4];
+  __shared__ alignas(16) u64 s_H0[16][4];
+  __shared__ i64 s_info[4];
+  __shared__ int s_abcd[4];
+  __shared__ unsigned long long s_eval;
+
+  const int n = args.n;
+  for (int i = threadIdx.x; i < n * 4; i += blockDim.x) {
+    s_pool[i] = reinterpret_cast<const u64*>(args.pool)[i];
+  }
+  if (threadIdx.x == 0) s_eval = 0;
+
+  DevCtl* ctl = args.ctl;
+  const i64 total4 = cf4(n);
+  u64 local_eval = 0;
+
+  for (;;) {
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      unsigned long long t = __hip_atomic_fetch_add(&ctl->queue, 1ULL,
+                                                    __ATOMIC_RELAXED,
+                                                    __HIP_MEMORY_SCOPE_AGENT);
+      s_info[0] = -1;
+      i64 qidx = static_cast<i64>(t);
+      if (qidx < total4 &&
+          __hip_atomic_load(&ctl->overflow, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT) == 0) {
+        // Decode quad (a,b,c,d).
+        int a = first_of_rank<cf4>(qidx, n, total4);
+        i64 rem = qidx - (total4 - cf4(n - a));
+        int m1 = n - a - 1;
+        int b0 = first_of_rank<cf3>(rem, m1, cf3(m1));
+        i64 rem2 = rem - (cf3(m1) - cf3(m1 - b0));
+        int c2, d2;
+        dev_decode_pair(rem2, m1 - b0 - 1, &c2, &d2);
+        int b = a + 1 + b0;
+        int c = b + 1 + c2;
+        int d = b + 1 + d2;
+        // Base rank of (a,b,c,d,d+1,d+2,d+3) in C(n,7): hockey-stick sums
+        // of the skipped blocks at each level.
+        i64 base = (cf7(n) - cf7(n - a)) + (cf6(n - a - 1) - cf6(n - b)) +
+                   (cf5(n - b - 1) - cf5(n - c)) + (cf4(n - c - 1) - cf4(n - d));
+        int m = n - 1 - d;
+        i64 ntrips = cf3(m);
+        i64 lo = args.begin > base ? args.begin - base : 0;
+        i64 hi = args.end - base < ntrips ? args.end - base : ntrips;
+        bool excl_prefix =
+            args.excl != 0 &&
+            (((a < 64) && ((args.excl >> a) & 1)) ||
+             ((b < 64) && ((args.excl >> b) & 1)) ||
+             ((c < 64) && ((args.excl >> c) & 1)) ||
+             ((d < 64) && ((args.excl >> d) & 1)));
+        if (base >= args.end) {
+          s_info[0] = -1;
+        } else if (lo >= hi || excl_prefix) {
+          s_info[0] = -2;
+        } else {
+          s_info[0] = base;
+          s_info[1] = lo;
+          s_info[2] = hi;
+          s_info[3] = m;
+          s_abcd[0] = a;
+          s_abcd[1] = b;
+          s_abcd[2] = c;
+          s_abcd[3] = d;
+        }
+      }
+    }
+    __syncthreads();
+    if (s_info[0] == -1) break;
+    if (s_info[0] == -2) continue;
+
+    const int a = s_abcd[0], b = s_abcd[1], c = s_abcd[2], d = s_abcd[3];
+    const int m = static_cast<int>(s_info[3]);
+    const i64 base = s_info[0];
+    const i64 lo = s_info[1], hi = s_info[2];
+
+    // 16 prefix cells x 4 words x {1,0} = 128 items.
+    if (threadIdx.x < 128) {
+      int u = threadIdx.x & 15;
+      int w = (threadIdx.x >> 4) & 3;
+      bool is1 = threadIdx.x >= 64;
+      u64 ca = (u & 8) ? s_pool[a * 4 + w] : ~s_pool[a * 4 + w];
+      u64 cb = (u & 4) ? s_pool[b * 4 + w] : ~s_pool[b * 4 + w];
+      u64 cc = (u & 2) ? s_pool[c * 4 + w] : ~s_pool[c * 4 + w];
+      u64 cd = (u & 1) ? s_pool[d * 4 + w] : ~s_pool[d * 4 + w];
+      u64 cell = ca & cb & cc & cd;
+      if (is1) {
+        s_H1[u][w] = cell & args.T1.w[w];
+      } else {
+        s_H0[u][w] = cell & args.T0.w[w];
+      }
+    }
+    __syncthreads();
+
+    int it = 0;
+    for (i64 q = lo + threadIdx.x; q < hi; q += blockDim.x) {
+      if (((it++) & 31) == 0) {
+        if ((!args.count_all && dev_abort(ctl)) ||
+            __hip_atomic_load(&ctl->overflow, __ATOMIC_RELAXED,
+                              __HIP_MEMORY_SCOPE_AGENT) != 0) {
+          break;
+        }
+      }
+      int e2, f2, g2;
+      dev_decode_triple(q, m, &e2, &f2, &g2);
+      int e = d + 1 + e2, f = d + 1 + f2, g = d + 1 + g2;
+      if (args.excl != 0) {
+        if ((e < 64 && ((args.excl >> e) & 1)) || (f < 64 && ((args.excl >> f) & 1)) ||
+            (g < 64 && ((args.excl >> g) & 1))) {
+          continue;
+        }
+      }
+      local_eval++;
+
+      u64 te_[4], tf_[4], tg_[4], nte_[4], ntf_[4], ntg_[4];
+#pragma unroll
+      for (int w = 0; w < 4; w++) {
+        te_[w] = s_pool[e * 4 + w];
+        tf_[w] = s_pool[f * 4 + w];
+        tg_[w] = s_pool[g * 4 + w];
+        nte_[w] = ~te_[w];
+        ntf_[w] = ~tf_[w];
+        ntg_[w] = ~tg_[w];
+      }
+
+      u64 p1[2] = {0, 0}, p0[2] = {0, 0};
+      bool ok = true;
+      for (int u = 0; u < 16 && ok; u++) {
+        // 8 sub-patterns of (e, f, g) per prefix cell.
+        u64 acc1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        u64 acc0[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+        for (int w = 0; w < 4; w++) {
+          const u64 h1 = s_H1[u][w];
+          const u64 h0 = s_H0[u][w];
+          const u64 e1 = h1 & te_[w], e0n = h1 & nte_[w];
+          const u64 z1 = h0 & te_[w], z0n = h0 & nte_[w];
+          const u64 ef11 = e1 & tf_[w], ef10 = e1 & ntf_[w];
+          const u64 ef01 = e0n & tf_[w], ef00 = e0n & ntf_[w];
+          const u64 zf11 = z1 & tf_[w], zf10 = z1 & ntf_[w];
+          const u64 zf01 = z0n & tf_[w], zf00 = z0n & ntf_[w];
+          acc1[7] |= ef11 & tg_[w];
+          acc1[6] |= ef11 & ntg_[w];
+          acc1[5] |= ef10 & tg_[w];
+          acc1[4] |= ef10 & ntg_[w];
+          acc1[3] |= ef01 & tg_[w];
+          acc1[2] |= ef01 & ntg_[w];
+          acc1[1] |= ef00 & tg_[w];
+          acc1[0] |= ef00 & ntg_[w];
+          acc0[7] |= zf11 & tg_[w];
+          acc0[6] |= zf11 & ntg_[w];
+          acc0[5] |= zf10 & tg_[w];
+          acc0[4] |= zf10 & ntg_[w];
+          acc0[3] |= zf01 & tg_[w];
+          acc0[2] |= zf01 & ntg_[w];
+          acc0[1] |= zf00 & tg_[w];
+          acc0[0] |= zf00 & ntg_[w];
+        }
+#pragma unroll
+        for (int p = 0; p < 8; p++) {
+          bool h1 = acc1[p] != 0, h0 = acc0[p] != 0;
+          if (h1 && h0) {
+            ok = false;
+            break;
+          }
+          int cell = (u << 3) | p;
+          if (h1) p1[cell >> 6] |= 1ULL << (cell & 63);
+          if (h0) p0[cell >> 6] |= 1ULL << (cell & 63);
+        }
+      }
+      if (!ok) continue;
+
+      // Survivor: append to hit buffer.
+      unsigned long long slot = __hip_atomic_fetch_add(&ctl->hit_count, 1ULL,
+                                                       __ATOMIC_RELAXED,
+                                                       __HIP_MEMORY_SCOPE_AGENT);
+      if (slot >= args.hit_cap) {
+        __hip_atomic_store(&ctl->overflow, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+      } else {
+        Hit7& h = args.hits[slot];
+        h.p1[0] = p1[0];
+        h.p1[1] = p1[1];
+        h.p0[0] = p0[0];
+        h.p0[1] = p0[1];
+        h.nums[0] = static_cast<u16>(a);
+        h.nums[1] = static_cast<u16>(b);
+        h.nums[2] = static_cast<u16>(c);
+        h.nums[3] = static_cast<u16>(d);
+        h.nums[4] = static_cast<u16>(e);
+        h.nums[5] = static_cast<u16>(f);
+        h.nums[6] = static_cast<u16>(g);
+      }
+    }
+  }
+
+  atomicAdd(&s_eval, static_cast<unsigned long long>(local_eval));
+  __syncthreads();
+  if (threadIdx.x == 0) atomicAdd(&ctl->evaluated, s_eval);
+}
+
+// ---------------------------------------------------------------------------
+// K4b — 7-LUT function assignment over filtered hits: one thread per
+// (hit, ordering); each runs the middle-function sweep + outer 2-coloring
+// (replacing the reference's 70 x 256 x 256 brute force, lut.c:416-484).
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(SCAN_BLOCK) k_scan7_assign(
+    const Hit7* hits, unsigned long long nhits, DevCtl* ctl, u64 seed) {
+  const u64 nitems = nhits * LUT7_NUM_ORDERINGS;
+  const u64 stride = static_cast<u64>(gridDim.x) * blockDim.x;
+  int tick = 0;
+  for (u64 item = blockIdx.x * static_cast<u64>(blockDim.x) + threadIdx.x;
+       item < nitems; item += stride) {
+    if (((tick++) & 15) == 0 && dev_abort(ctl)) return;
+    const Hit7& h = hits[item / LUT7_NUM_ORDERINGS];
+    int oidx = static_cast<int>(item % LUT7_NUM_ORDERINGS);
+    u8 ord[7];
+    lut7_ordering(oidx, ord);
+    u8 fo, fm, fi;
+    if (lut7_solve_ordering(h.p1, h.p0, ord, dev_rnd(seed, item), &fo, &fm, &fi)) {
+      u16 res[10];
+      res[0] = fo;
+      res[1] = fm;
+      res[2] = fi;
+      for (int j = 0; j < 7; j++) res[3 + j] = h.nums[ord[j]];
+      dev_publish(ctl, res);
+      return;
+    }
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// GpuEngine host wrapper.
+// ---------------------------------------------------------------------------
+
+struct GpuEngine::Impl {
+  int device = 0;
+  hipStream_t stream = nullptr;
+  ttable* d_pool = nullptr;
+  DevCtl* d_ctl = nullptr;
+  Hit7* d_hits = nullptr;
+  u64 hit_cap = 0;
+  DevCtl* h_ctl = nullptr;  // pinned staging
+  std::string name;
+
+  ~Impl() {
+    if (d_pool != nullptr) (void)hipFree(d_pool);
+    if (d_ctl != nullptr) (void)hipFree(d_ctl);
+    if (d_hits != nullptr) (void)hipFree(d_hits);
+    if (h_ctl != nullptr) (void)hipHostFree(h_ctl);
+    if (stream != nullptr) (void)hipStreamDestroy(stream);
+  }
+};
+
+bool gpu_available() { return gpu_count() > 0; }
+
+int gpu_count() {
+  int count = 0;
+  if (hipGetDeviceCount(&count) != hipSuccess) return 0;
+  return count;
+}
+
+std::unique_ptr<GpuEngine> GpuEngine::create(int device, std::string* err) {
+  int count = 0;
+  hipError_t e = hipGetDeviceCount(&count);
+  if (e != hipSuccess || count == 0) {
+    if (err != nullptr) {
+      *err = e == hipSuccess ? "no HIP devices visible" : hipGetErrorString(e);
+    }
+    return nullptr;
+  }
+  try {
+    auto impl = std::make_unique<Impl>();
+    if (device >= 0) {
+      SBG_HIP_CHECK(hipSetDevice(device));
+      impl->device = device;
+    } else {
+      SBG_HIP_CHECK(hipGetDevice(&impl->device));
+    }
+    hipDeviceProp_t prop;
+    SBG_HIP_CHECK(hipGetDeviceProperties(&prop, impl->device));
+    impl->name = prop.name;
+    SBG_HIP_CHECK(hipStreamCreate(&impl->stream));
+    SBG_HIP_CHECK(hipMalloc(&impl->d_pool, sizeof(ttable) * MAX_GATES));
+    SBG_HIP_CHECK(hipMalloc(&impl->d_ctl, sizeof(DevCtl)));
+    SBG_HIP_CHECK(hipHostMalloc(&impl->h_ctl, sizeof(DevCtl)));
+    // Hit buffer for the 7-LUT frontier: default 16M hits (768 MB) per
+    // chunk; overridable for memory-constrained runs.
+    const char* cap_env = std::getenv("SBOXGATES_HIT_CAP");
+    impl->hit_cap = cap_env != nullptr ? std::strtoull(cap_env, nullptr, 10)
+                                       : (1ULL << 24);
+    SBG_HIP_CHECK(hipMalloc(&impl->d_hits, sizeof(Hit7) * impl->hit_cap));
+    return std::unique_ptr<GpuEngine>(new GpuEngine(impl.release()));
+  } catch (const std::exception& ex) {
+    if (err != nullptr) *err = ex.what();
+    return nullptr;
+  }
+}
+
+GpuEngine::~GpuEngine() { delete impl_; }
+
+int GpuEngine::device() const { return impl_->device; }
+std::string GpuEngine::device_name() const { return impl_->name; }
+
+ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
+  ScanResult out;
+  const i64 total = n_choose_k(rq.n, k);
+  if (begin >= total) return out;
+  if (end > total) end = total;
+  if (begin >= end) return out;
+
+  Impl* im = impl_;
+  SBG_HIP_CHECK(hipSetDevice(im->device));
+
+  // Upload the pool and reset the control block.
+  SBG_HIP_CHECK(hipMemcpyAsync(im->d_pool, rq.tables, sizeof(ttable) * rq.n,
+                               hipMemcpyHostToDevice, im->stream));
+  std::memset(im->h_ctl, 0, sizeof(DevCtl));
+
+  ScanArgs args;
+  args.pool = im->d_pool;
+  args.ctl = im->d_ctl;
+  args.hits = im->d_hits;
+  args.hit_cap = im->hit_cap;
+  args.T1 = rq.target & rq.mask;
+  args.T0 = ~rq.target & rq.mask;
+  args.n = rq.n;
+  args.begin = begin;
+  args.end = end;
+  args.excl = rq.excl_low64;
+  args.seed = rq.seed;
+  args.count_all = rq.count_all ? 1 : 0;
+
+  if (k == 3) {
+    SBG_HIP_CHECK(hipMemcpyAsync(im->d_ctl, im->h_ctl, sizeof(DevCtl),
+                                 hipMemcpyHostToDevice, im->stream));
+    i64 range = end - begin;
+    int grid = static_cast<int>(std::min<i64>((range + SCAN_BLOCK - 1) / SCAN_BLOCK,
+                                              4096));
+    hipLaunchKernelGGL(k_scan3, dim3(grid), dim3(SCAN_BLOCK), 0, im->stream, args);
+    SBG_HIP_CHECK(hipGetLastError());
+    SBG_HIP_CHECK(hipMemcpyAsync(im->h_ctl, im->d_ctl, sizeof(DevCtl),
+                                 hipMemcpyDeviceToHost, im->stream));
+    SBG_HIP_CHECK(hipStreamSynchronize(im->stream));
+  } else if (k == 5) {
+    // Seed the triple queue at the triple containing `begin`.
+    gatenum first[5];
+    nth_combination(begin, rq.n, 5, 0, first);
+    i64 t_begin = combination_rank(first, 3, rq.n);
+    im->h_ctl->queue = static_cast<unsigned long long>(t_begin);
+    SBG_HIP_CHECK(hipMemcpyAsync(im->d_ctl, im->h_ctl, sizeof(DevCtl),
+                                 hipMemcpyHostToDevice, im->stream));
+    int grid = 2048;
+    hipLaunchKernelGGL(k_scan5, dim3(grid), dim3(SCAN_BLOCK), 0, im->stream, args);
+    SBG_HIP_CHECK(hipGetLastError());
+    SBG_HIP_CHECK(hipMemcpyAsync(im->h_ctl, im->d_ctl, sizeof(DevCtl),
+                                 hipMemcpyDeviceToHost, im->stream));
+    SBG_HIP_CHECK(hipStreamSynchronize(im->stream));
+  } else if (k == 7) {
+    // Filter + assign, with overflow-driven range splitting.
+    i64 lo = begin;
+    while (lo < end) {
+      i64 hi = end;
+      for (;;) {
+        std::memset(im->h_ctl, 0, sizeof(DevCtl));
+        gatenum first[7];
+        nth_combination(lo, rq.n, 7, 0, first);
+        im->h_ctl->queue =
+            static_cast<unsigned long long>(combination_rank(first, 4, rq.n));
+        SBG_HIP_CHECK(hipMemcpyAsync(im->d_ctl, im->h_ctl, sizeof(DevCtl),
+                                     hipMemcpyHostToDevice, im->stream));
+        ScanArgs a2 = args;
+        a2.begin = lo;
+        a2.end = hi;
+        hipLaunchKernelGGL(k_scan7_filter, dim3(2048), dim3(SCAN_BLOCK), 0,
+                           im->stream, a2);
+        SBG_HIP_CHECK(hipGetLastError());
+        SBG_HIP_CHECK(hipMemcpyAsync(im->h_ctl, im->d_ctl, sizeof(DevCtl),
+                                     hipMemcpyDeviceToHost, im->stream));
+        SBG_HIP_CHECK(hipStreamSynchronize(im->stream));
+        if (im->h_ctl->overflow == 0) break;
+        // Too many feasible combinations for the buffer: halve the range.
+        hi = lo + (hi - lo) / 2;
+        if (hi <= lo + 1) {
+          throw std::runtime_error("7-LUT hit buffer too small for one combo");
+        }
+      }
+      out.evaluated += im->h_ctl->evaluated;
+      unsigned long long nhits = im->h_ctl->hit_count;
+      if (nhits > im->hit_cap) nhits = im->hit_cap;
+      if (nhits > 0 && !rq.count_all) {
+        u64 items = nhits * LUT7_NUM_ORDERINGS;
+        int grid = static_cast<int>(
+            std::min<u64>((items + SCAN_BLOCK - 1) / SCAN_BLOCK, 4096));
+        hipLaunchKernelGGL(k_scan7_assign, dim3(grid), dim3(SCAN_BLOCK), 0,
+                           im->stream, im->d_hits, nhits, im->d_ctl, rq.seed);
+        SBG_HIP_CHECK(hipGetLastError());
+        SBG_HIP_CHECK(hipMemcpyAsync(im->h_ctl, im->d_ctl, sizeof(DevCtl),
+                                     hipMemcpyDeviceToHost, im->stream));
+        SBG_HIP_CHECK(hipStreamSynchronize(im->stream));
+        if (im->h_ctl->found != 0) {
+          out.found = true;
+          std::memcpy(out.res, im->h_ctl->res, sizeof(out.res));
+          return out;
+        }
+      }
+      lo = hi;
+    }
+    return out;
+  } else {
+    throw std::runtime_error("GpuEngine::scan: bad k");
+  }
+
+  out.evaluated = im->h_ctl->evaluated;
+  if (im->h_ctl->found != 0) {
+    out.found = true;
+    std::memcpy(out.res, im->h_ctl->res, sizeof(out.res));
+  }
+  return out;
+}
+
+}  // namespace sbg

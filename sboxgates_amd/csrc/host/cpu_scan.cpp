@@ -1,0 +1,175 @@
+// cpu_scan.cpp — CPU implementations of the 3/5/7-LUT combination scans
+// (no-GPU execution path + oracle for the HIP kernels), plus the naive
+// reference-style checkers used as independent test oracles.
+
+#include "sbg/scan.hpp"
+
+#include "sbg/comb.hpp"
+#include "sbg/lutcover.hpp"
+#include "sbg/rng.hpp"
+
+namespace sbg {
+
+static inline bool excluded(const ScanRequest& rq, const gatenum* nums, int k) {
+  if (rq.excl_low64 == 0) return false;
+  for (int i = 0; i < k; i++) {
+    if (nums[i] < 64 && (rq.excl_low64 >> nums[i]) & 1) return true;
+  }
+  return false;
+}
+
+ScanResult cpu_scan3(const ScanRequest& rq, i64 begin, i64 end) {
+  ScanResult out;
+  const i64 total = n_choose_k(rq.n, 3);
+  if (begin >= total) return out;
+  if (end > total) end = total;
+
+  const ttable T1 = rq.target & rq.mask;
+  const ttable T0 = ~rq.target & rq.mask;
+
+  gatenum nums[3];
+  nth_combination(begin, rq.n, 3, 0, nums);
+  for (i64 i = begin; i < end; i++) {
+    out.evaluated++;
+    // NOTE: the reference's 3-LUT scan does not reject inbits combinations
+    // (lut.c:501-523) — parity kept: no exclusion here.
+    u32 p1, p0;
+    if (lut3_p_masks(rq.tables[nums[0]], rq.tables[nums[1]], rq.tables[nums[2]],
+                     T1, T0, &p1, &p0)) {
+      u8 func = lut3_function_from_p(p1, p0, hash_mix64(rq.seed ^ static_cast<u64>(i)));
+      if (!rq.count_all) {
+        out.found = true;
+        out.res[0] = func;
+        out.res[1] = nums[0];
+        out.res[2] = nums[1];
+        out.res[3] = nums[2];
+        return out;
+      }
+    }
+    next_combination(nums, 3, rq.n);
+  }
+  return out;
+}
+
+ScanResult cpu_scan5(const ScanRequest& rq, i64 begin, i64 end) {
+  ScanResult out;
+  const i64 total = n_choose_k(rq.n, 5);
+  if (begin >= total) return out;
+  if (end > total) end = total;
+
+  const ttable T1 = rq.target & rq.mask;
+  const ttable T0 = ~rq.target & rq.mask;
+
+  gatenum nums[5];
+  nth_combination(begin, rq.n, 5, 0, nums);
+  ttable tt[5];
+  for (int j = 0; j < 5; j++) tt[j] = rq.tables[nums[j]];
+
+  for (i64 i = begin; i < end; i++) {
+    out.evaluated++;
+    if (!excluded(rq, nums, 5)) {
+      u32 p1, p0;
+      if (lut5_p_masks(tt, T1, T0, &p1, &p0)) {
+        u8 fo, fi;
+        int split;
+        u64 rnd = hash_mix64(rq.seed ^ static_cast<u64>(i));
+        if (lut5_solve_from_p(p1, p0, rnd, &fo, &fi, &split) && !rq.count_all) {
+          const u8* sp = SPLITS5[split];
+          out.found = true;
+          out.res[0] = fo;
+          out.res[1] = fi;
+          for (int j = 0; j < 3; j++) out.res[2 + j] = nums[sp[j]];
+          out.res[5] = nums[sp[3]];
+          out.res[6] = nums[sp[4]];
+          return out;
+        }
+      }
+    }
+    next_combination(nums, 5, rq.n);
+    for (int j = 0; j < 5; j++) tt[j] = rq.tables[nums[j]];
+  }
+  return out;
+}
+
+ScanResult cpu_scan7(const ScanRequest& rq, i64 begin, i64 end) {
+  ScanResult out;
+  const i64 total = n_choose_k(rq.n, 7);
+  if (begin >= total) return out;
+  if (end > total) end = total;
+
+  const ttable T1 = rq.target & rq.mask;
+  const ttable T0 = ~rq.target & rq.mask;
+
+  gatenum nums[7];
+  nth_combination(begin, rq.n, 7, 0, nums);
+  ttable tt[7];
+  for (int j = 0; j < 7; j++) tt[j] = rq.tables[nums[j]];
+
+  for (i64 i = begin; i < end; i++) {
+    out.evaluated++;
+    if (!excluded(rq, nums, 7)) {
+      u64 p1[2], p0[2];
+      if (lut7_p_masks(tt, T1, T0, p1, p0)) {
+        u64 rnd = hash_mix64(rq.seed ^ static_cast<u64>(i));
+        for (int o = 0; o < LUT7_NUM_ORDERINGS; o++) {
+          u8 ord[7];
+          lut7_ordering(o, ord);
+          u8 fo, fm, fi;
+          if (lut7_solve_ordering(p1, p0, ord, rnd, &fo, &fm, &fi)) {
+            if (!rq.count_all) {
+              out.found = true;
+              out.res[0] = fo;
+              out.res[1] = fm;
+              out.res[2] = fi;
+              for (int j = 0; j < 7; j++) out.res[3 + j] = nums[ord[j]];
+              return out;
+            }
+            break;
+          }
+        }
+      }
+    }
+    next_combination(nums, 7, rq.n);
+    for (int j = 0; j < 7; j++) tt[j] = rq.tables[nums[j]];
+  }
+  return out;
+}
+
+// --- Naive oracles (fresh implementations of the reference semantics) ---
+
+bool naive_check_n_lut_possible(int num, const ttable& target, const ttable& mask,
+                                const ttable* tables) {
+  // A k-LUT exists iff no input-pattern cell mixes masked target-1 and
+  // target-0 positions.
+  for (int c = 0; c < (1 << num); c++) {
+    ttable cell = tt_ones_table();
+    for (int j = 0; j < num; j++) {
+      cell &= ((c >> (num - 1 - j)) & 1) ? tables[j] : ~tables[j];
+    }
+    bool has1 = tt_any(cell & target & mask);
+    bool has0 = tt_any(cell & ~target & mask);
+    if (has1 && has0) return false;
+  }
+  return true;
+}
+
+bool naive_get_lut_function(const ttable& a, const ttable& b, const ttable& c,
+                            const ttable& target, const ttable& mask, u8* func) {
+  // Per-position constraint propagation over all 256 positions.
+  u8 f = 0, set = 0;
+  for (int pos = 0; pos < 256; pos++) {
+    if (!tt_get_bit(mask, pos)) continue;
+    int p = (tt_get_bit(a, pos) << 2) | (tt_get_bit(b, pos) << 1) | tt_get_bit(c, pos);
+    u8 want = static_cast<u8>(tt_get_bit(target, pos));
+    if (set & (1u << p)) {
+      if (((f >> p) & 1) != want) return false;
+    } else {
+      set |= 1u << p;
+      f |= static_cast<u8>(want << p);
+    }
+  }
+  *func = f;
+  return true;
+}
+
+}  // namespace sbg
