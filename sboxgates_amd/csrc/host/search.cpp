@@ -6,6 +6,7 @@
 #include <cassert>
 #include <climits>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <stdexcept>
 
@@ -335,9 +336,16 @@ bool Engine::distributed_lut_body(const WorkMsg& w, u16 res[10], bool* found5) {
   }
 
   *found5 = false;
-  // 5-LUT phase (chunk size: fixed constant so every rank agrees).
-  constexpr u64 CHUNK5 = 1ULL << 30;
-  constexpr u64 CHUNK7 = 1ULL << 30;
+  // 5/7-LUT chunk sizes: fixed constants (env-overridable for tests) so
+  // every rank derives the same chunk count from the broadcast state.
+  static const u64 CHUNK5 = [] {
+    const char* s = std::getenv("SBOXGATES_CHUNK5");
+    return s != nullptr ? std::strtoull(s, nullptr, 10) : (1ULL << 30);
+  }();
+  static const u64 CHUNK7 = [] {
+    const char* s = std::getenv("SBOXGATES_CHUNK7");
+    return s != nullptr ? std::strtoull(s, nullptr, 10) : (1ULL << 30);
+  }();
   if (w.st.num_gates >= 5) {
     if (dist_scan_chunked(5, rq, CHUNK5, res)) {
       *found5 = true;
