@@ -460,6 +460,7 @@ PYBIND11_MODULE(_core, m) {
         rq.excl_low64 = 0;
         rq.seed = seed;
         rq.count_all = count_all;
+        if (k == 4) rq.matcher = e.matcher3();
         ScanResult r = e.scan(k, rq, begin, end);
         std::vector<int> res(r.res, r.res + 10);
         return py::make_tuple(r.found, res, r.evaluated);

@@ -164,6 +164,7 @@ struct ScanArgs {
   DevCtl* ctl;
   Hit7* hits;          // K7 only
   u64 hit_cap;         // K7 only
+  const Avail3Matcher* matcher;  // k=4 only
   ttable T1, T0;
   int n;
   i64 begin, end;
@@ -222,6 +223,102 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan3(ScanArgs args) {
     }
   }
   // Per-block evaluated reduction.
+  __shared__ unsigned long long s_eval;
+  if (threadIdx.x == 0) s_eval = 0;
+  __syncthreads();
+  atomicAdd(&s_eval, static_cast<unsigned long long>(local_eval));
+  __syncthreads();
+  if (threadIdx.x == 0) atomicAdd(&ctl->evaluated, s_eval);
+}
+
+// ---------------------------------------------------------------------------
+// K1c — gate-mode step-4 triple scan: triples realized by an available
+// composed 3-input function in one of 6 argument orders. The per-candidate
+// function search is one bitmap probe per order (matcher precomputed on the
+// host), replacing the reference's 256-function x 4-order truth-table loop
+// (sboxgates.c:392-435).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ u8 dev_permute_cells8(u8 m, int s0, int s1, int s2) {
+  u8 out = 0;
+#pragma unroll
+  for (int c = 0; c < 8; c++) {
+    if (!((m >> c) & 1)) continue;
+    int v[3] = {(c >> 2) & 1, (c >> 1) & 1, c & 1};
+    out |= static_cast<u8>(1u << ((v[s0] << 2) | (v[s1] << 1) | v[s2]));
+  }
+  return out;
+}
+
+__constant__ int c_perm6[6][3] = {{0, 1, 2}, {0, 2, 1}, {1, 0, 2},
+                                  {1, 2, 0}, {2, 0, 1}, {2, 1, 0}};
+
+__global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * 4];
+  __shared__ alignas(16) u8 s_bitmap[256 * 256 / 8];
+  __shared__ u8 s_funs[256];
+  __shared__ int s_count;
+  const int n = args.n;
+  for (int i = threadIdx.x; i < n * 4; i += blockDim.x) {
+    s_pool[i] = reinterpret_cast<const u64*>(args.pool)[i];
+  }
+  for (int i = threadIdx.x; i < 256 * 256 / 8 / 8; i += blockDim.x) {
+    reinterpret_cast<u64*>(s_bitmap)[i] =
+        reinterpret_cast<const u64*>(args.matcher->bitmap)[i];
+  }
+  for (int i = threadIdx.x; i < 256 / 8; i += blockDim.x) {
+    reinterpret_cast<u64*>(s_funs)[i] =
+        reinterpret_cast<const u64*>(args.matcher->funs)[i];
+  }
+  if (threadIdx.x == 0) s_count = args.matcher->count;
+  __syncthreads();
+
+  DevCtl* ctl = args.ctl;
+  const i64 total3 = cf3(n);
+  u64 local_eval = 0;
+  const i64 stride = static_cast<i64>(gridDim.x) * blockDim.x;
+  i64 idx = args.begin + blockIdx.x * static_cast<i64>(blockDim.x) + threadIdx.x;
+  int tick = 0;
+
+  for (; idx < args.end; idx += stride) {
+    if (((tick++) & 255) == 0 && !args.count_all && dev_abort(ctl)) break;
+    int a = first_of_rank<cf3>(idx, n, total3);
+    i64 rem = idx - (total3 - cf3(n - a));
+    int b2, c2;
+    dev_decode_pair(rem, n - a - 1, &b2, &c2);
+    int b = a + 1 + b2, c = a + 1 + c2;
+
+    local_eval++;
+    const ttable* ta = reinterpret_cast<const ttable*>(&s_pool[a * 4]);
+    const ttable* tb = reinterpret_cast<const ttable*>(&s_pool[b * 4]);
+    const ttable* tc = reinterpret_cast<const ttable*>(&s_pool[c * 4]);
+    u32 p1, p0;
+    if (!lut3_p_masks(*ta, *tb, *tc, args.T1, args.T0, &p1, &p0)) continue;
+    if (args.count_all) continue;
+    const u8 req1 = static_cast<u8>(p1);
+    const u8 care = static_cast<u8>(p1 | p0);
+    for (int perm = 0; perm < 6; perm++) {
+      const u8 r = dev_permute_cells8(req1, c_perm6[perm][0], c_perm6[perm][1],
+                                      c_perm6[perm][2]);
+      const u8 cr = dev_permute_cells8(care, c_perm6[perm][0], c_perm6[perm][1],
+                                       c_perm6[perm][2]);
+      const int bidx = cr * 256 + r;
+      if (!((s_bitmap[bidx >> 3] >> (bidx & 7)) & 1)) continue;
+      for (int f = 0; f < s_count; f++) {
+        if ((s_funs[f] & cr) == r) {
+          u16 res[10] = {};
+          res[0] = static_cast<u16>(f);
+          res[1] = static_cast<u16>(perm);
+          res[2] = static_cast<u16>(a);
+          res[3] = static_cast<u16>(b);
+          res[4] = static_cast<u16>(c);
+          dev_publish(ctl, res);
+          break;
+        }
+      }
+      break;  // bitmap said a function exists; it was found and published
+    }
+    if (dev_abort(ctl) && !args.count_all) break;
+  }
   __shared__ unsigned long long s_eval;
   if (threadIdx.x == 0) s_eval = 0;
   __syncthreads();
@@ -668,12 +765,14 @@ struct GpuEngine::Impl {
   ttable* d_pool = nullptr;
   DevCtl* d_ctl = nullptr;
   Hit7* d_hits = nullptr;
+  Avail3Matcher* d_matcher = nullptr;
   u64 hit_cap = 0;
   DevCtl* h_ctl = nullptr;  // pinned staging
   std::string name;
 
   ~Impl() {
     if (d_pool != nullptr) (void)hipFree(d_pool);
+    if (d_matcher != nullptr) (void)hipFree(d_matcher);
     if (d_ctl != nullptr) (void)hipFree(d_ctl);
     if (d_hits != nullptr) (void)hipFree(d_hits);
     if (h_ctl != nullptr) (void)hipHostFree(h_ctl);
@@ -712,6 +811,7 @@ std::unique_ptr<GpuEngine> GpuEngine::create(int device, std::string* err) {
     SBG_HIP_CHECK(hipStreamCreate(&impl->stream));
     SBG_HIP_CHECK(hipMalloc(&impl->d_pool, sizeof(ttable) * MAX_GATES));
     SBG_HIP_CHECK(hipMalloc(&impl->d_ctl, sizeof(DevCtl)));
+    SBG_HIP_CHECK(hipMalloc(&impl->d_matcher, sizeof(Avail3Matcher)));
     SBG_HIP_CHECK(hipHostMalloc(&impl->h_ctl, sizeof(DevCtl)));
     // Hit buffer for the 7-LUT frontier: default 16M hits (768 MB) per
     // chunk; overridable for memory-constrained runs.
@@ -733,7 +833,7 @@ std::string GpuEngine::device_name() const { return impl_->name; }
 
 ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   ScanResult out;
-  const i64 total = n_choose_k(rq.n, k);
+  const i64 total = n_choose_k(rq.n, k == 4 ? 3 : k);
   if (begin >= total) return out;
   if (end > total) end = total;
   if (begin >= end) return out;
@@ -751,6 +851,7 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   args.ctl = im->d_ctl;
   args.hits = im->d_hits;
   args.hit_cap = im->hit_cap;
+  args.matcher = im->d_matcher;
   args.T1 = rq.target & rq.mask;
   args.T0 = ~rq.target & rq.mask;
   args.n = rq.n;
@@ -760,13 +861,23 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   args.seed = rq.seed;
   args.count_all = rq.count_all ? 1 : 0;
 
-  if (k == 3) {
+  if (k == 3 || k == 4) {
+    if (k == 4) {
+      if (rq.matcher == nullptr) throw std::runtime_error("scan4 needs matcher");
+      SBG_HIP_CHECK(hipMemcpyAsync(im->d_matcher, rq.matcher,
+                                   sizeof(Avail3Matcher), hipMemcpyHostToDevice,
+                                   im->stream));
+    }
     SBG_HIP_CHECK(hipMemcpyAsync(im->d_ctl, im->h_ctl, sizeof(DevCtl),
                                  hipMemcpyHostToDevice, im->stream));
     i64 range = end - begin;
     int grid = static_cast<int>(std::min<i64>((range + SCAN_BLOCK - 1) / SCAN_BLOCK,
                                               4096));
-    hipLaunchKernelGGL(k_scan3, dim3(grid), dim3(SCAN_BLOCK), 0, im->stream, args);
+    if (k == 3) {
+      hipLaunchKernelGGL(k_scan3, dim3(grid), dim3(SCAN_BLOCK), 0, im->stream, args);
+    } else {
+      hipLaunchKernelGGL(k_scan4, dim3(grid), dim3(SCAN_BLOCK), 0, im->stream, args);
+    }
     SBG_HIP_CHECK(hipGetLastError());
     SBG_HIP_CHECK(hipMemcpyAsync(im->h_ctl, im->d_ctl, sizeof(DevCtl),
                                  hipMemcpyDeviceToHost, im->stream));

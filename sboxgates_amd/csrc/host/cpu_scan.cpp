@@ -4,6 +4,8 @@
 
 #include "sbg/scan.hpp"
 
+#include <cstring>
+
 #include "sbg/comb.hpp"
 #include "sbg/lutcover.hpp"
 #include "sbg/rng.hpp"
@@ -131,6 +133,82 @@ ScanResult cpu_scan7(const ScanRequest& rq, i64 begin, i64 end) {
     }
     next_combination(nums, 7, rq.n);
     for (int j = 0; j < 7; j++) tt[j] = rq.tables[nums[j]];
+  }
+  return out;
+}
+
+// --- Gate-mode step-4 triple scan (k = 4) ---
+
+const int TRIPLE_PERMS6[6][3] = {{0, 1, 2}, {0, 2, 1}, {1, 0, 2},
+                                 {1, 2, 0}, {2, 0, 1}, {2, 1, 0}};
+
+u8 permute_cells8_host(u8 m, const int* sel) {
+  u8 out = 0;
+  for (int c = 0; c < 8; c++) {
+    if (!((m >> c) & 1)) continue;
+    int v[3] = {(c >> 2) & 1, (c >> 1) & 1, c & 1};
+    out |= static_cast<u8>(1u << ((v[sel[0]] << 2) | (v[sel[1]] << 1) | v[sel[2]]));
+  }
+  return out;
+}
+
+void build_avail3_matcher(const u8* funs, const u8* costs, int count,
+                          Avail3Matcher* out) {
+  std::memset(out->bitmap, 0, sizeof(out->bitmap));
+  out->count = count;
+  for (int i = 0; i < count; i++) {
+    out->funs[i] = funs[i];
+    out->cost[i] = costs != nullptr ? costs[i] : 1;
+  }
+  // bitmap[care][req1] = exists f with (f & care) == req1. Enumerate per
+  // function: for each care, the satisfied req1 is f & care.
+  for (int i = 0; i < count; i++) {
+    for (int care = 0; care < 256; care++) {
+      int req1 = funs[i] & care;
+      int idx = care * 256 + req1;
+      out->bitmap[idx >> 3] |= static_cast<u8>(1u << (idx & 7));
+    }
+  }
+}
+
+ScanResult cpu_scan4(const ScanRequest& rq, i64 begin, i64 end) {
+  ScanResult out;
+  const i64 total = n_choose_k(rq.n, 3);
+  if (begin >= total) return out;
+  if (end > total) end = total;
+  const Avail3Matcher* M = rq.matcher;
+
+  const ttable T1 = rq.target & rq.mask;
+  const ttable T0 = ~rq.target & rq.mask;
+
+  gatenum nums[3];
+  nth_combination(begin, rq.n, 3, 0, nums);
+  for (i64 i = begin; i < end; i++) {
+    out.evaluated++;
+    u32 p1, p0;
+    if (lut3_p_masks(rq.tables[nums[0]], rq.tables[nums[1]], rq.tables[nums[2]],
+                     T1, T0, &p1, &p0)) {
+      const u8 req1 = static_cast<u8>(p1);
+      const u8 care = static_cast<u8>(p1 | p0);
+      for (int perm = 0; perm < 6 && !rq.count_all; perm++) {
+        const u8 r = permute_cells8_host(req1, TRIPLE_PERMS6[perm]);
+        const u8 c = permute_cells8_host(care, TRIPLE_PERMS6[perm]);
+        const int idx = c * 256 + r;
+        if (!((M->bitmap[idx >> 3] >> (idx & 7)) & 1)) continue;
+        for (int f = 0; f < M->count; f++) {
+          if ((M->funs[f] & c) == r) {
+            out.found = true;
+            out.res[0] = static_cast<u16>(f);
+            out.res[1] = static_cast<u16>(perm);
+            out.res[2] = nums[0];
+            out.res[3] = nums[1];
+            out.res[4] = nums[2];
+            return out;
+          }
+        }
+      }
+    }
+    next_combination(nums, 3, rq.n);
   }
   return out;
 }

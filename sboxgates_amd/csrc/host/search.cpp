@@ -201,22 +201,6 @@ static inline u8 swap_pair_patterns(u8 m) {
   return static_cast<u8>((m & 0x9) | ((m & 2) << 1) | ((m & 4) >> 1));
 }
 
-// Permute an 8-bit cell mask for a triple: out bit (v[s0]<<2|v[s1]<<1|v[s2])
-// = in bit (v0<<2|v1<<1|v2).
-static inline u8 permute_cells8(u8 m, const int* sel) {
-  u8 out = 0;
-  for (int c = 0; c < 8; c++) {
-    if (!((m >> c) & 1)) continue;
-    int v0 = (c >> 2) & 1, v1 = (c >> 1) & 1, v2 = c & 1;
-    int v[3] = {v0, v1, v2};
-    out |= static_cast<u8>(1u << ((v[sel[0]] << 2) | (v[sel[1]] << 1) | v[sel[2]]));
-  }
-  return out;
-}
-
-static const int TRIPLE_PERMS[6][3] = {{0, 1, 2}, {0, 2, 1}, {1, 0, 2},
-                                       {1, 2, 0}, {2, 0, 1}, {2, 1, 0}};
-
 // ---------------------------------------------------------------------------
 // Engine
 // ---------------------------------------------------------------------------
@@ -237,6 +221,21 @@ Engine::Engine(const options& opt, DistCtx* ctx)
 }
 
 Engine::~Engine() = default;
+
+const Avail3Matcher* Engine::matcher3() {
+  if (matcher_ == nullptr) {
+    matcher_ = std::make_unique<Avail3Matcher>();
+    u8 funs[256], costs[256];
+    for (int i = 0; i < opt_.num_avail_3; i++) {
+      const boolfunc& f = opt_.avail_3[i];
+      funs[i] = f.fun;
+      costs[i] = static_cast<u8>(2 + (f.not_a ? 1 : 0) + (f.not_b ? 1 : 0) +
+                                 (f.not_c ? 1 : 0) + (f.not_out ? 1 : 0));
+    }
+    build_avail3_matcher(funs, costs, opt_.num_avail_3, matcher_.get());
+  }
+  return matcher_.get();
+}
 
 bool Engine::gpu_active() const { return gpu_ != nullptr; }
 
@@ -265,13 +264,15 @@ ScanResult Engine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
     stats_.cpu_scans += 1;
     switch (k) {
       case 3: r = cpu_scan3(rq, begin, end); break;
+      case 4: r = cpu_scan4(rq, begin, end); break;
       case 5: r = cpu_scan5(rq, begin, end); break;
       case 7: r = cpu_scan7(rq, begin, end); break;
       default: throw std::runtime_error("bad scan k");
     }
   }
   switch (k) {
-    case 3: stats_.candidates3 += r.evaluated; break;
+    case 3:
+    case 4: stats_.candidates3 += r.evaluated; break;
     case 5: stats_.candidates5 += r.evaluated; break;
     default: stats_.candidates7 += r.evaluated; break;
   }
@@ -573,42 +574,35 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
     }
 
     // Step 4b: triples realized by an available composed 3-input function
-    // (sboxgates.c:388-435). Cell-requirement screen + byte-pattern match;
-    // all 6 argument orders (improvement over the reference's 4).
+    // (sboxgates.c:388-435) — the gate-mode hot loop, run as a k=4 scan
+    // (GPU kernel k_scan4 on large pools): cell-requirement screen + one
+    // matcher-bitmap probe per argument order; all 6 orders (improvement
+    // over the reference's 4 orders with mis-indexed commutativity gates).
     if (!check_num_gates_possible(st, 3, 2 * sat_metric_of(AND) + sat_metric_of(NOT),
                                   opt_.metric)) {
       return NO_GATE;
     }
-    for (int i = 0; i < st->num_gates; i++) {
-      const gatenum gi = gate_order[i];
-      const ttable& ti = st->gates[gi].table;
-      for (int k = i + 1; k < st->num_gates; k++) {
-        const gatenum gk = gate_order[k];
-        const ttable& tk = st->gates[gk].table;
-        for (int m = k + 1; m < st->num_gates; m++) {
-          const gatenum gm = gate_order[m];
-          const ttable& tm = st->gates[gm].table;
-          u32 p1, p0;
-          if (!lut3_p_masks(ti, tk, tm, T1, T0, &p1, &p0)) continue;
-          const u8 req1 = static_cast<u8>(p1);
-          const u8 care = static_cast<u8>(p1 | p0);
-          const gatenum ids[3] = {gi, gk, gm};
-          for (int perm = 0; perm < 6; perm++) {
-            // req in the basis where argument j is ids[sel[j]]: the inverse
-            // permutation maps canonical cells to permuted patterns.
-            const int* sel = TRIPLE_PERMS[perm];
-            u8 req1p = permute_cells8(req1, sel);
-            u8 carep = permute_cells8(care, sel);
-            for (int p = 0; p < opt_.num_avail_3; p++) {
-              if ((opt_.avail_3[p].fun & carep) == req1p) {
-                return assert_ret(
-                    add_boolfunc_3(st, opt_.avail_3[p], ids[sel[0]], ids[sel[1]],
-                                   ids[sel[2]], opt_.metric),
-                    target, st, mask, "step4b");
-              }
-            }
-          }
-        }
+    if (opt_.num_avail_3 > 0 && st->num_gates >= 3) {
+      static thread_local std::vector<ttable> pool4;
+      pool4.resize(st->num_gates);
+      for (int i = 0; i < st->num_gates; i++) pool4[i] = st->gates[i].table;
+      ScanRequest rq;
+      rq.tables = pool4.data();
+      rq.n = st->num_gates;
+      rq.target = target;
+      rq.mask = mask;
+      rq.excl_low64 = 0;
+      rq.seed = rng_.next();
+      rq.count_all = false;
+      rq.matcher = matcher3();
+      ScanResult r = scan(4, rq, 0, n_choose_k(rq.n, 3));
+      if (r.found) {
+        const boolfunc& f = opt_.avail_3[r.res[0]];
+        const int* sel = TRIPLE_PERMS6[r.res[1]];
+        const gatenum ids[3] = {r.res[2], r.res[3], r.res[4]};
+        return assert_ret(add_boolfunc_3(st, f, ids[sel[0]], ids[sel[1]],
+                                         ids[sel[2]], opt_.metric),
+                          target, st, mask, "step4b");
       }
     }
   }

@@ -70,6 +70,8 @@ class Engine {
   ScanResult scan(int k, const ScanRequest& rq, i64 begin, i64 end);
 
   bool gpu_active() const;
+  // Cached step-4 matcher built from this engine's avail_3 vocabulary.
+  const Avail3Matcher* matcher3();
   const SearchStats& stats() const { return stats_; }
   const std::vector<std::string>& saved_files() const { return saved_files_; }
   options& opt() { return opt_; }
@@ -92,6 +94,7 @@ class Engine {
   int num_outputs_ = 0;
   ttable g_target_[8] = {};
   std::unique_ptr<GpuEngine> gpu_;
+  std::unique_ptr<Avail3Matcher> matcher_;
   SearchStats stats_;
   std::vector<std::string> saved_files_;
 };

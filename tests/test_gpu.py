@@ -172,3 +172,57 @@ def test_scan5_throughput_sane(engines):
     print(f"gpu {gpu_rate:.3g} cand/s, cpu {cpu_rate:.3g} cand/s, "
           f"speedup {gpu_rate / cpu_rate:.1f}x")
     assert gpu_rate > 20 * cpu_rate
+
+
+def test_scan4_gpu_matches_cpu(engines):
+    """Gate-mode step-4 triple kernel vs CPU path: counts + planted hits."""
+    import random
+    from sboxgates_amd.ops import function_lists, gen_lut_ttable, tt_eq_mask
+    gpu = make_engine(seed=1, gpu="force", save_states=False, try_nots=True)
+    cpu = make_engine(seed=1, gpu="off", save_states=False, try_nots=True)
+    sbox, n = models.load("rijndael")
+    gpu.set_sbox(sbox, n)
+    cpu.set_sbox(sbox, n)
+    st = gpu.initial_state()
+    st.grow_pool_random(120, 0xFEED)
+    mask = mask_for_inputs(8)
+    total = n_choose_k(120, 3)
+    # Count parity.
+    _, _, ev_g = gpu.scan_pool(4, st, gpu.target(0), mask, 0, total,
+                               count_all=True)
+    _, _, ev_c = cpu.scan_pool(4, st, cpu.target(0), mask, 0, total,
+                               count_all=True)
+    assert ev_g == ev_c == total
+    # Planted composed-function targets must be found and verify.
+    _, _, threes = function_lists(2 + 64 + 128, True)
+    rng = random.Random(3)
+    perms = [(0, 1, 2), (0, 2, 1), (1, 0, 2), (1, 2, 0), (2, 0, 1), (2, 1, 0)]
+    for trial in range(5):
+        f = rng.choice(threes)
+        ids = sorted(rng.sample(range(120), 3))
+        target = gen_lut_ttable(f["fun"], st.gate(ids[0])["table"],
+                                st.gate(ids[1])["table"], st.gate(ids[2])["table"])
+        found, res, _ = gpu.scan_pool(4, st, target, mask, 0, total)
+        assert found, trial
+        sel = perms[res[1]]
+        gids = [res[2], res[3], res[4]]
+        got = gen_lut_ttable(threes[res[0]]["fun"],
+                             st.gate(gids[sel[0]])["table"],
+                             st.gate(gids[sel[1]])["table"],
+                             st.gate(gids[sel[2]])["table"])
+        assert tt_eq_mask(target, got, mask)
+
+
+def test_rijndael_gate_mode_bit0_gpu():
+    """AES S-box output bit 0, 2-input {AND,OR,XOR} gate set, single GPU —
+    BASELINE config 2's single-output core, with the step-4 triple scan on
+    the k_scan4 kernel."""
+    eng = make_engine(seed=11, gpu="force", save_states=False)
+    sbox, n = models.load("rijndael")
+    eng.set_sbox(sbox, n)
+    st = eng.initial_state()
+    out = eng.create_circuit(st, eng.target(0), mask_for_inputs(n))
+    assert out >= 0
+    st.set_output(0, out)
+    assert validate_circuit(st, sbox, n, bit=0)
+    assert eng.stats()["gpu_scans"] > 0

@@ -206,3 +206,62 @@ class TestScans:
                              st.gate(res[7])["table"], st.gate(res[8])["table"])
         g_i = gen_lut_ttable(res[2], g_o, g_m, st.gate(res[9])["table"])
         assert tt_eq_mask(t, g_i, mask)
+
+
+class TestScan4:
+    """Gate-mode step-4 triple scan (matcher-based)."""
+
+    def make(self, pool=30, bitfield=2 + 64 + 128, try_nots=False):
+        sbox, n = models.load("rijndael")
+        eng = make_engine(seed=1, gpu="off", save_states=False,
+                          gate_bitfield=bitfield, try_nots=try_nots)
+        eng.set_sbox(sbox, n)
+        st = eng.initial_state()
+        st.grow_pool_random(pool, 0xD00D)
+        return eng, st, mask_for_inputs(8)
+
+    def test_counts(self):
+        eng, st, mask = self.make()
+        total = n_choose_k(st.num_gates, 3)
+        found, res, ev = eng.scan_pool(4, st, eng.target(0), mask, 0, total,
+                                       count_all=True)
+        assert ev == total
+
+    def test_planted_composed_function_found(self):
+        from sboxgates_amd.ops import function_lists, gen_lut_ttable, tt_eq_mask
+        import random
+        rng = random.Random(5)
+        eng, st, mask = self.make(try_nots=True)
+        _, _, threes = function_lists(2 + 64 + 128, True)
+        for trial in range(10):
+            f = rng.choice(threes)
+            ids = sorted(rng.sample(range(st.num_gates), 3))
+            target = gen_lut_ttable(f["fun"], st.gate(ids[0])["table"],
+                                    st.gate(ids[1])["table"],
+                                    st.gate(ids[2])["table"])
+            found, res, ev = eng.scan_pool(4, st, target, mask, 0,
+                                           n_choose_k(st.num_gates, 3))
+            assert found, trial
+            # Verify: avail fun at res[0] applied in order res[1] to the
+            # found triple matches the target under the mask.
+            perms = [(0, 1, 2), (0, 2, 1), (1, 0, 2), (1, 2, 0), (2, 0, 1),
+                     (2, 1, 0)]
+            sel = perms[res[1]]
+            gids = [res[2], res[3], res[4]]
+            fun = threes[res[0]]["fun"]
+            got = gen_lut_ttable(fun, st.gate(gids[sel[0]])["table"],
+                                 st.gate(gids[sel[1]])["table"],
+                                 st.gate(gids[sel[2]])["table"])
+            assert tt_eq_mask(target, got, mask)
+
+    def test_scan4_matches_gate_search(self):
+        """The scan-based step 4b must still let gate-mode searches finish
+        (covered end-to-end by des tests); here: a target with no matching
+        triple must not be 'found'."""
+        eng, st, mask = self.make(pool=12, bitfield=2)  # AND only: tiny closure
+        target = eng.target(3)
+        found, res, ev = eng.scan_pool(4, st, target, mask, 0,
+                                       n_choose_k(st.num_gates, 3))
+        # With only AND compositions over a random pool, realizing an AES
+        # output bit exactly is essentially impossible.
+        assert not found
