@@ -6,23 +6,29 @@
 // (-c/-d mutually exclusive, -l/-s mutually exclusive). MI355X additions:
 // --convert-hip, --seed N, --cpu / --gpu, --output-dir DIR.
 //
-// Unlike the reference there is no MPI launcher: a single process drives
-// one GPU (or the CPU path). Multi-GPU runs go through the Python layer
-// (sboxgates_amd.parallel, torchrun one process per GPU over RCCL).
+// Unlike the reference there is no MPI launcher: --gpus N drives N devices
+// from ONE process (one host thread + one engine per GPU, shared-memory
+// coordination over the same chunked protocol). The Python layer
+// (sboxgates_amd.parallel, torchrun one process per GPU over RCCL) covers
+// multi-process deployment.
 
 #include <getopt.h>
 
+#include <atomic>
 #include <climits>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
 #include <string>
+#include <thread>
+#include <vector>
 
 #include "sbg/codegen.hpp"
 #include "sbg/gpu.hpp"
 #include "sbg/options.hpp"
 #include "sbg/sboxio.hpp"
 #include "sbg/search.hpp"
+#include "sbg/threads_ctx.hpp"
 #include "sbg/xmlio.hpp"
 
 namespace {
@@ -59,7 +65,9 @@ void print_help(const char* prog) {
       "      --seed=N                  Deterministic RNG seed.\n"
       "      --cpu                     Disable the GPU path.\n"
       "      --gpu                     Require the GPU path (fail if no device).\n"
-      "      --output-dir=DIR          Directory for XML checkpoint files.\n\n"
+      "      --output-dir=DIR          Directory for XML checkpoint files.\n"
+      "      --gpus=N                  Drive N GPUs from this process (one\n"
+      "                                worker thread per device).\n\n"
       "  -?, --help                    Give this help list.\n"
       "  -V, --version                 Print program version.\n",
       prog);
@@ -80,7 +88,8 @@ int main(int argc, char** argv) {
   sbg::options opt;
   opt.set_avail_gates(sbg::DEFAULT_GATE_BITFIELD);
 
-  enum { OPT_HIP = 1000, OPT_SEED, OPT_CPU, OPT_GPU, OPT_OUTDIR, OPT_HELP };
+  enum { OPT_HIP = 1000, OPT_SEED, OPT_CPU, OPT_GPU, OPT_OUTDIR, OPT_HELP,
+         OPT_GPUS };
   static const struct option long_opts[] = {
       {"available-gates", required_argument, nullptr, 'a'},
       {"convert-c", no_argument, nullptr, 'c'},
@@ -98,6 +107,7 @@ int main(int argc, char** argv) {
       {"cpu", no_argument, nullptr, OPT_CPU},
       {"gpu", no_argument, nullptr, OPT_GPU},
       {"output-dir", required_argument, nullptr, OPT_OUTDIR},
+      {"gpus", required_argument, nullptr, OPT_GPUS},
       {"help", no_argument, nullptr, OPT_HELP},
       {"version", no_argument, nullptr, 'V'},
       {nullptr, 0, nullptr, 0}};
@@ -147,6 +157,11 @@ int main(int argc, char** argv) {
       case OPT_CPU: opt.gpu = sbg::GPU_OFF; break;
       case OPT_GPU: opt.gpu = sbg::GPU_FORCE; break;
       case OPT_OUTDIR: opt.output_dir = optarg; break;
+      case OPT_GPUS:
+        v = std::strtol(optarg, &endptr, 10);
+        if (*endptr != '\0' || v < 1 || v > 64) return fail("Bad --gpus value", optarg);
+        opt.num_gpus = static_cast<int>(v);
+        break;
       case 'V': std::printf("%s\n", kVersion); return 0;
       case OPT_HELP: print_help(argv[0]); return 0;
       default: return 1;
@@ -198,7 +213,28 @@ int main(int argc, char** argv) {
       }
     }
 
-    sbg::Engine engine(opt);
+    sbg::ThreadGroup group(opt.num_gpus);
+    std::vector<std::thread> workers;
+    std::atomic<bool> worker_failed{false};
+    const int dev_count = sbg::gpu_count();
+    for (int r = 1; r < opt.num_gpus; r++) {
+      workers.emplace_back([&, r] {
+        try {
+          sbg::options wopt = opt;
+          wopt.gpu_device = dev_count > 0 ? r % dev_count : -1;
+          sbg::Engine we(wopt, group.ctx(r));
+          we.set_sbox(sbox, static_cast<int>(num_inputs));
+          we.worker_loop();
+        } catch (const std::exception& e) {
+          std::fprintf(stderr, "worker %d: %s\n", r, e.what());
+          worker_failed = true;
+          std::exit(1);  // protocol is broken; do not deadlock rank 0
+        }
+      });
+    }
+    sbg::options opt0 = opt;
+    opt0.gpu_device = dev_count > 0 ? 0 : -1;
+    sbg::Engine engine(opt0, group.ctx(0));
     engine.set_sbox(sbox, static_cast<int>(num_inputs));
 
     if (opt.verbosity >= 1) {
@@ -240,6 +276,9 @@ int main(int argc, char** argv) {
     } else {
       engine.generate_graph(st);
     }
+    engine.stop_workers();
+    for (auto& t : workers) t.join();
+    if (worker_failed.load()) return 1;
   } catch (const std::exception& e) {
     std::fprintf(stderr, "%s\n", e.what());
     return 1;

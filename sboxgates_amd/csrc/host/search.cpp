@@ -213,7 +213,7 @@ Engine::Engine(const options& opt, DistCtx* ctx)
   }
   if (opt_.gpu != GPU_OFF) {
     std::string err;
-    gpu_ = GpuEngine::create(-1, &err);
+    gpu_ = GpuEngine::create(opt_.gpu_device, &err);
     if (gpu_ == nullptr && opt_.gpu == GPU_FORCE) {
       throw std::runtime_error("sboxgates: GPU required but unavailable: " + err);
     }

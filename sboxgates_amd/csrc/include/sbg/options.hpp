@@ -40,6 +40,8 @@ struct options {
   bool seeded = false;      // --seed given: deterministic RNG
   u64 seed = 0;
   gpu_mode_t gpu = GPU_AUTO;
+  int gpu_device = -1;      // HIP device index; -1 = current device
+  int num_gpus = 1;         // CLI --gpus: in-process devices (threads)
   std::string output_dir;   // where XML checkpoints are written ("" = CWD)
   bool save_states = true;  // library callers may disable checkpoint writes
 

@@ -133,3 +133,16 @@ def test_identity_sbox(tmp_path):
     r = run(["-o", "0", "--seed", "3", "--cpu",
              os.path.join(SBOX, "identity.txt")], cwd=str(tmp_path))
     assert r.returncode == 0
+
+
+def test_multi_device_threads(tmp_path):
+    """--gpus N: in-process SPMD over worker threads (CPU path here; on a
+    GPU node each thread drives its own device)."""
+    r = run(["-l", "-o", "0", "--seed", "4", "--cpu", "--gpus", "3", des()],
+            cwd=str(tmp_path), timeout=240)
+    assert r.returncode == 0, r.stderr
+    assert glob.glob(os.path.join(str(tmp_path), "1-*.xml"))
+
+
+def test_bad_gpus_value():
+    assert run(["--gpus", "0", des()]).returncode != 0
