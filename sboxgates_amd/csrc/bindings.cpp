@@ -445,6 +445,9 @@ PYBIND11_MODULE(_core, m) {
         d["candidates7"] = e.stats().candidates7;
         d["gpu_scans"] = e.stats().gpu_scans;
         d["cpu_scans"] = e.stats().cpu_scans;
+        d["scan_seconds3"] = e.stats().scan_seconds3;
+        d["scan_seconds5"] = e.stats().scan_seconds5;
+        d["scan_seconds7"] = e.stats().scan_seconds7;
         return d;
       })
       .def("scan_pool", [](Engine& e, int k, const state& st, py::bytes target,

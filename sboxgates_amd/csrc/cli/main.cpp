@@ -279,6 +279,20 @@ int main(int argc, char** argv) {
     engine.stop_workers();
     for (auto& t : workers) t.join();
     if (worker_failed.load()) return 1;
+    if (opt.verbosity >= 1) {
+      const auto& st_ = engine.stats();
+      auto rate = [](sbg::u64 c, double s) { return s > 0 ? c / s : 0.0; };
+      std::printf("Scan totals: 3-in %llu (%.3g/s), 5LUT %llu (%.3g/s), "
+                  "7LUT %llu (%.3g/s); %llu GPU / %llu CPU scans\n",
+                  (unsigned long long)st_.candidates3,
+                  rate(st_.candidates3, st_.scan_seconds3),
+                  (unsigned long long)st_.candidates5,
+                  rate(st_.candidates5, st_.scan_seconds5),
+                  (unsigned long long)st_.candidates7,
+                  rate(st_.candidates7, st_.scan_seconds7),
+                  (unsigned long long)st_.gpu_scans,
+                  (unsigned long long)st_.cpu_scans);
+    }
   } catch (const std::exception& e) {
     std::fprintf(stderr, "%s\n", e.what());
     return 1;

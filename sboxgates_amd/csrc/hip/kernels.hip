@@ -175,17 +175,30 @@ struct ScanArgs {
 
 constexpr int SCAN_BLOCK = 256;
 
+// LDS pool stride: 6 u64 (48 B) per gate instead of 4 (32 B). A 32 B
+// stride puts ds_read_b128 lane groups on 8-way-conflicting banks when
+// lanes read consecutive gate ids; 48 B (12 dwords, 16 B aligned) spreads
+// a 16-lane group over 16 distinct banks (measured: SQ_LDS_BANK_CONFLICT
+// was 21% of LDS cycles at stride 32).
+constexpr int PSTR = 6;
+
+__device__ __forceinline__ void load_pool_lds(u64* s_pool, const ttable* pool,
+                                              int n) {
+  const u64* src = reinterpret_cast<const u64*>(pool);
+  for (int i = threadIdx.x; i < n * 4; i += blockDim.x) {
+    s_pool[(i >> 2) * PSTR + (i & 3)] = src[i];
+  }
+}
+
 // ---------------------------------------------------------------------------
 // K2 — 3-LUT scan. Grid-stride over combination ranks; per-thread decode by
 // binary search on closed-form binomials; cells evaluated directly (the
 // reference's serial rank-0 loop, lut.c:501-523).
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(SCAN_BLOCK) k_scan3(ScanArgs args) {
-  __shared__ alignas(16) u64 s_pool[MAX_GATES * 4];
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * PSTR];
   const int n = args.n;
-  for (int i = threadIdx.x; i < n * 4; i += blockDim.x) {
-    s_pool[i] = reinterpret_cast<const u64*>(args.pool)[i];
-  }
+  load_pool_lds(s_pool, args.pool, n);
   __syncthreads();
 
   DevCtl* ctl = args.ctl;
@@ -205,9 +218,9 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan3(ScanArgs args) {
     int b = a + 1 + b2, c = a + 1 + c2;
 
     local_eval++;
-    const ttable* ta = reinterpret_cast<const ttable*>(&s_pool[a * 4]);
-    const ttable* tb = reinterpret_cast<const ttable*>(&s_pool[b * 4]);
-    const ttable* tc = reinterpret_cast<const ttable*>(&s_pool[c * 4]);
+    const ttable* ta = reinterpret_cast<const ttable*>(&s_pool[a * PSTR]);
+    const ttable* tb = reinterpret_cast<const ttable*>(&s_pool[b * PSTR]);
+    const ttable* tc = reinterpret_cast<const ttable*>(&s_pool[c * PSTR]);
     u32 p1, p0;
     if (lut3_p_masks(*ta, *tb, *tc, args.T1, args.T0, &p1, &p0)) {
       if (!args.count_all) {
@@ -253,14 +266,12 @@ __constant__ int c_perm6[6][3] = {{0, 1, 2}, {0, 2, 1}, {1, 0, 2},
                                   {1, 2, 0}, {2, 0, 1}, {2, 1, 0}};
 
 __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
-  __shared__ alignas(16) u64 s_pool[MAX_GATES * 4];
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * PSTR];
   __shared__ alignas(16) u8 s_bitmap[256 * 256 / 8];
   __shared__ u8 s_funs[256];
   __shared__ int s_count;
   const int n = args.n;
-  for (int i = threadIdx.x; i < n * 4; i += blockDim.x) {
-    s_pool[i] = reinterpret_cast<const u64*>(args.pool)[i];
-  }
+  load_pool_lds(s_pool, args.pool, n);
   for (int i = threadIdx.x; i < 256 * 256 / 8 / 8; i += blockDim.x) {
     reinterpret_cast<u64*>(s_bitmap)[i] =
         reinterpret_cast<const u64*>(args.matcher->bitmap)[i];
@@ -288,9 +299,9 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
     int b = a + 1 + b2, c = a + 1 + c2;
 
     local_eval++;
-    const ttable* ta = reinterpret_cast<const ttable*>(&s_pool[a * 4]);
-    const ttable* tb = reinterpret_cast<const ttable*>(&s_pool[b * 4]);
-    const ttable* tc = reinterpret_cast<const ttable*>(&s_pool[c * 4]);
+    const ttable* ta = reinterpret_cast<const ttable*>(&s_pool[a * PSTR]);
+    const ttable* tb = reinterpret_cast<const ttable*>(&s_pool[b * PSTR]);
+    const ttable* tc = reinterpret_cast<const ttable*>(&s_pool[c * PSTR]);
     u32 p1, p0;
     if (!lut3_p_masks(*ta, *tb, *tc, args.T1, args.T0, &p1, &p0)) continue;
     if (args.count_all) continue;
@@ -335,99 +346,130 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
 // 10 x 256-function brute force, lut.c:174-246).
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(SCAN_BLOCK) k_scan5(ScanArgs args) {
-  __shared__ alignas(16) u64 s_pool[MAX_GATES * 4];
-  __shared__ alignas(16) u64 s_H1[8][4];
-  __shared__ alignas(16) u64 s_H0[8][4];
-  __shared__ i64 s_info[4];  // base rank, pair lo, pair hi, m
-  __shared__ int s_abc[4];   // a, b, c, skip-flag
+  // Triple batching: one dequeue + one cooperative prefix-cell build + one
+  // barrier per TB triples (measured at TB=1: 48% of wave time parked on
+  // the per-triple barriers; batching amortizes them 8x).
+  constexpr int TB = 8;
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * PSTR];
+  __shared__ alignas(16) u64 s_H1[TB][8][4];
+  __shared__ alignas(16) u64 s_H0[TB][8][4];
+  __shared__ i64 s_base[TB];     // flat-rank base of each triple
+  __shared__ i64 s_lo[TB], s_prefix[TB + 1];
+  __shared__ int s_m[TB], s_c[TB];
+  __shared__ u16 s_abc[TB][3];
+  __shared__ int s_stop;         // all triples past range end
   __shared__ unsigned long long s_eval;
 
   const int n = args.n;
-  for (int i = threadIdx.x; i < n * 4; i += blockDim.x) {
-    s_pool[i] = reinterpret_cast<const u64*>(args.pool)[i];
-  }
+  load_pool_lds(s_pool, args.pool, n);
   if (threadIdx.x == 0) s_eval = 0;
 
   DevCtl* ctl = args.ctl;
   const i64 total3 = cf3(n);
-  // Triple range covered by [begin, end): ranks group by leading triple.
-  // Host passes the first triple's index via queue start = 0 and we add
-  // t_begin here, computed from begin.
   u64 local_eval = 0;
 
   for (;;) {
     __syncthreads();
     if (threadIdx.x == 0) {
-      // Dequeue the next triple (device-scope).
-      unsigned long long t = __hip_atomic_fetch_add(&ctl->queue, 1ULL,
-                                                    __ATOMIC_RELAXED,
-                                                    __HIP_MEMORY_SCOPE_AGENT);
-      s_info[0] = -1;
-      if (!(!args.count_all && dev_abort(ctl))) {
-        i64 tidx = static_cast<i64>(t);
-        if (tidx < total3) {
-          int a = first_of_rank<cf3>(tidx, n, total3);
-          i64 rem = tidx - (total3 - cf3(n - a));
-          int b2, c2;
-          dev_decode_pair(rem, n - a - 1, &b2, &c2);
-          int b = a + 1 + b2, c = a + 1 + c2;
-          // Base rank of (a,b,c,c+1,c+2) in C(n,5) via hockey-stick sums.
-          i64 base = (cf5(n) - cf5(n - a)) + (cf4(n - a - 1) - cf4(n - b)) +
-                     (cf3(n - b - 1) - cf3(n - c));
-          int m = n - 1 - c;  // remaining gates c+1..n-1
-          i64 npairs = cf2(m);
-          i64 lo = args.begin > base ? args.begin - base : 0;
-          i64 hi = args.end - base < npairs ? args.end - base : npairs;
-          bool excl_prefix =
-              args.excl != 0 &&
-              (((a < 64) && ((args.excl >> a) & 1)) ||
-               ((b < 64) && ((args.excl >> b) & 1)) ||
-               ((c < 64) && ((args.excl >> c) & 1)));
-          if (base >= args.end) {
-            s_info[0] = -1;  // past the range: stop
-          } else if (lo >= hi || excl_prefix) {
-            s_info[0] = -2;  // nothing to do for this triple; next
-          } else {
-            s_info[0] = base;
-            s_info[1] = lo;
-            s_info[2] = hi;
-            s_info[3] = m;
-            s_abc[0] = a;
-            s_abc[1] = b;
-            s_abc[2] = c;
-          }
-        }
+      s_stop = (!args.count_all && dev_abort(ctl)) ? 1 : 0;
+      if (!s_stop) {
+        unsigned long long t0 = __hip_atomic_fetch_add(
+            &ctl->queue, static_cast<unsigned long long>(TB), __ATOMIC_RELAXED,
+            __HIP_MEMORY_SCOPE_AGENT);
+        s_base[0] = static_cast<i64>(t0);  // reuse slot to pass the batch base
       }
     }
     __syncthreads();
-    if (s_info[0] == -1) break;
-    if (s_info[0] == -2) continue;
+    if (s_stop) break;
+    const i64 batch_base = s_base[0];
+    __syncthreads();
 
-    const int a = s_abc[0], b = s_abc[1], c = s_abc[2];
-    const int m = static_cast<int>(s_info[3]);
-    const i64 base = s_info[0];
-    const i64 lo = s_info[1], hi = s_info[2];
+    // Threads 0..TB-1 decode one triple each and compute its pair window.
+    if (threadIdx.x < TB) {
+      const int t = threadIdx.x;
+      const i64 tidx = batch_base + t;
+      i64 base = -1, lo = 0, hi = 0;
+      int m = 0, a = 0, b = 0, c = 0;
+      if (tidx < total3) {
+        a = first_of_rank<cf3>(tidx, n, total3);
+        i64 rem = tidx - (total3 - cf3(n - a));
+        int b2, c2;
+        dev_decode_pair(rem, n - a - 1, &b2, &c2);
+        b = a + 1 + b2;
+        c = a + 1 + c2;
+        base = (cf5(n) - cf5(n - a)) + (cf4(n - a - 1) - cf4(n - b)) +
+               (cf3(n - b - 1) - cf3(n - c));
+        m = n - 1 - c;
+        i64 npairs = cf2(m);
+        lo = args.begin > base ? args.begin - base : 0;
+        hi = args.end - base < npairs ? args.end - base : npairs;
+        bool excl_prefix =
+            args.excl != 0 &&
+            (((a < 64) && ((args.excl >> a) & 1)) ||
+             ((b < 64) && ((args.excl >> b) & 1)) ||
+             ((c < 64) && ((args.excl >> c) & 1)));
+        if (base >= args.end || lo >= hi || excl_prefix) {
+          lo = hi = 0;
+        }
+      }
+      s_base[t] = base;
+      s_lo[t] = lo;
+      s_m[t] = m;
+      s_c[t] = c;
+      s_abc[t][0] = static_cast<u16>(a);
+      s_abc[t][1] = static_cast<u16>(b);
+      s_abc[t][2] = static_cast<u16>(c);
+      s_prefix[t + 1] = hi - lo;  // count, turned into prefix sums below
+      if (t == 0) s_prefix[0] = 0;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      // Prefix-sum the TB pair counts. Termination: triple flat-rank bases
+      // increase with the triple index, so once the batch's first triple
+      // is past the range end (or past the triple space), every later
+      // batch is too.
+      s_stop = (batch_base >= total3 || s_base[0] >= args.end) ? 1 : 0;
+      for (int t = 0; t < TB; t++) s_prefix[t + 1] += s_prefix[t];
+    }
+    __syncthreads();
+    if (s_stop) break;
+    const i64 total_pairs = s_prefix[TB];
+    if (total_pairs == 0) continue;
 
-    // Cooperative prefix-cell build: 64 items = (u, word, which).
-    if (threadIdx.x < 64) {
-      int u = threadIdx.x & 7;
-      int w = (threadIdx.x >> 3) & 3;
-      bool is1 = threadIdx.x >= 32;
-      u64 ca = (u & 4) ? s_pool[a * 4 + w] : ~s_pool[a * 4 + w];
-      u64 cb = (u & 2) ? s_pool[b * 4 + w] : ~s_pool[b * 4 + w];
-      u64 cc = (u & 1) ? s_pool[c * 4 + w] : ~s_pool[c * 4 + w];
-      u64 cell = ca & cb & cc;
-      if (is1) {
-        s_H1[u][w] = cell & args.T1.w[w];
-      } else {
-        s_H0[u][w] = cell & args.T0.w[w];
+    // Cooperative prefix-cell build: TB x 8 cells x 4 words x {1,0} = 512
+    // items, two per thread.
+    {
+      const int half = TB * 8 * 4;  // 256 at TB=8
+      for (int item = threadIdx.x; item < 2 * half; item += blockDim.x) {
+        const bool is1 = item < half;
+        const int it = is1 ? item : item - half;
+        const int t = it >> 5;          // /32
+        const int u = (it >> 2) & 7;
+        const int w = it & 3;
+        if (s_base[t] < 0 || s_prefix[t + 1] == s_prefix[t]) continue;
+        const int a = s_abc[t][0], b = s_abc[t][1], c = s_abc[t][2];
+        u64 ca = (u & 4) ? s_pool[a * PSTR + w] : ~s_pool[a * PSTR + w];
+        u64 cb = (u & 2) ? s_pool[b * PSTR + w] : ~s_pool[b * PSTR + w];
+        u64 cc = (u & 1) ? s_pool[c * PSTR + w] : ~s_pool[c * PSTR + w];
+        u64 cell = ca & cb & cc;
+        if (is1) {
+          s_H1[t][u][w] = cell & args.T1.w[w];
+        } else {
+          s_H0[t][u][w] = cell & args.T0.w[w];
+        }
       }
     }
     __syncthreads();
 
     int it = 0;
-    for (i64 q = lo + threadIdx.x; q < hi; q += blockDim.x) {
+    for (i64 idx = threadIdx.x; idx < total_pairs; idx += blockDim.x) {
       if (((it++) & 31) == 0 && !args.count_all && dev_abort(ctl)) break;
+      // Locate the triple for this flat index (TB-entry scan).
+      int t = 0;
+      while (s_prefix[t + 1] <= idx) t++;
+      const i64 q = s_lo[t] + (idx - s_prefix[t]);
+      const int m = s_m[t];
+      const int c = s_c[t];
       int d2, e2;
       dev_decode_pair(q, m, &d2, &e2);
       int d = c + 1 + d2, e = c + 1 + e2;
@@ -438,8 +480,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan5(ScanArgs args) {
       }
       local_eval++;
 
-      const u64* td = &s_pool[d * 4];
-      const u64* te = &s_pool[e * 4];
+      const u64* td = &s_pool[d * PSTR];
+      const u64* te = &s_pool[e * PSTR];
       u64 td_[4], te_[4], ntd_[4], nte_[4];
 #pragma unroll
       for (int w = 0; w < 4; w++) {
@@ -457,8 +499,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan5(ScanArgs args) {
         u64 r11_0 = 0, r10_0 = 0, r01_0 = 0, r00_0 = 0;
 #pragma unroll
         for (int w = 0; w < 4; w++) {
-          const u64 h1 = s_H1[u][w];
-          const u64 h0 = s_H0[u][w];
+          const u64 h1 = s_H1[t][u][w];
+          const u64 h0 = s_H0[t][u][w];
           const u64 a1 = h1 & td_[w];
           const u64 na1 = h1 & ntd_[w];
           const u64 a0 = h0 & td_[w];
@@ -489,11 +531,11 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan5(ScanArgs args) {
 
       u8 fo, fi;
       int split;
-      if (lut5_solve_from_p(p1, p0, dev_rnd(args.seed, base + q), &fo, &fi, &split)) {
+      if (lut5_solve_from_p(p1, p0, dev_rnd(args.seed, s_base[t] + q), &fo, &fi,
+                            &split)) {
         if (!args.count_all) {
-          const u16 nums[5] = {static_cast<u16>(a), static_cast<u16>(b),
-                               static_cast<u16>(c), static_cast<u16>(d),
-                               static_cast<u16>(e)};
+          const u16 nums[5] = {s_abc[t][0], s_abc[t][1], s_abc[t][2],
+                               static_cast<u16>(d), static_cast<u16>(e)};
           const u8* sp = SPLITS5[split];
           u16 res[10] = {};
           res[0] = fo;

@@ -4,6 +4,7 @@
 #include "sbg/search.hpp"
 
 #include <cassert>
+#include <chrono>
 #include <climits>
 #include <cstdio>
 #include <cstdlib>
@@ -256,6 +257,7 @@ ScanResult Engine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   constexpr i64 GPU_MIN_CANDIDATES = 1 << 16;
   bool use_gpu = gpu_ != nullptr &&
                  (opt_.gpu == GPU_FORCE || end - begin >= GPU_MIN_CANDIDATES);
+  const auto t0 = std::chrono::steady_clock::now();
   ScanResult r;
   if (use_gpu) {
     stats_.gpu_scans += 1;
@@ -270,11 +272,27 @@ ScanResult Engine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
       default: throw std::runtime_error("bad scan k");
     }
   }
+  const double dt =
+      std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
   switch (k) {
     case 3:
-    case 4: stats_.candidates3 += r.evaluated; break;
-    case 5: stats_.candidates5 += r.evaluated; break;
-    default: stats_.candidates7 += r.evaluated; break;
+    case 4:
+      stats_.candidates3 += r.evaluated;
+      stats_.scan_seconds3 += dt;
+      break;
+    case 5:
+      stats_.candidates5 += r.evaluated;
+      stats_.scan_seconds5 += dt;
+      break;
+    default:
+      stats_.candidates7 += r.evaluated;
+      stats_.scan_seconds7 += dt;
+      break;
+  }
+  if (opt_.verbosity >= 3) {
+    std::printf("[%4d] scan%d: %lld candidates in %.3fs (%.3g cand/s)%s\n",
+                ctx_->rank(), k, static_cast<long long>(r.evaluated), dt,
+                dt > 0 ? r.evaluated / dt : 0.0, r.found ? " HIT" : "");
   }
   return r;
 }

@@ -33,6 +33,9 @@ struct SearchStats {
   u64 candidates7 = 0;
   u64 gpu_scans = 0;
   u64 cpu_scans = 0;
+  double scan_seconds3 = 0;  // wall time inside 3-input scans (incl. k=4)
+  double scan_seconds5 = 0;
+  double scan_seconds7 = 0;
 };
 
 class Engine {
