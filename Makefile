@@ -61,3 +61,19 @@ clean:
 -include $(BUILD)/*.d
 
 .PHONY: all ext cli clean
+
+# Sanitizer self-tests: host engine under ASan+UBSan / TSan (g++; the HIP
+# runtime is stubbed out — CPU scan paths only).
+SAN_SRCS := $(HOST_SRCS) $(SRC)/test/gpu_stub.cpp $(SRC)/test/selftest.cpp
+
+asan: | $(BUILD)
+	g++ -O1 -g -std=c++20 -fsanitize=address,undefined -fno-omit-frame-pointer \
+	  $(INC) $(SAN_SRCS) -o $(BUILD)/selftest_asan -lpthread
+	$(BUILD)/selftest_asan
+
+tsan: | $(BUILD)
+	g++ -O1 -g -std=c++20 -fsanitize=thread $(INC) $(SAN_SRCS) \
+	  -o $(BUILD)/selftest_tsan -lpthread
+	$(BUILD)/selftest_tsan
+
+.PHONY: asan tsan

@@ -366,6 +366,7 @@ bool Engine::distributed_lut_body(const WorkMsg& w, u16 res[10], bool* found5) {
     return s != nullptr ? std::strtoull(s, nullptr, 10) : (1ULL << 30);
   }();
   if (w.st.num_gates >= 5) {
+    if (w.verbosity >= 2 && ctx_->rank() == 0) std::printf("[   0] Search 5.\n");
     if (dist_scan_chunked(5, rq, CHUNK5, res)) {
       *found5 = true;
       return true;
@@ -374,6 +375,7 @@ bool Engine::distributed_lut_body(const WorkMsg& w, u16 res[10], bool* found5) {
   // 7-LUT phase; the gate check is symmetric (same broadcast state).
   if (!check_num_gates_possible(&w.st, 3, 0, METRIC_GATES)) return false;
   if (w.st.num_gates >= 7) {
+    if (w.verbosity >= 2 && ctx_->rank() == 0) std::printf("[   0] Search 7.\n");
     return dist_scan_chunked(7, rq, CHUNK7, res);
   }
   return false;
