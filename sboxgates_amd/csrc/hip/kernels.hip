@@ -345,7 +345,38 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
 // 2-coloring decomposition solver on survivors (replacing the reference's
 // 10 x 256-function brute force, lut.c:174-246).
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(SCAN_BLOCK) k_scan5(ScanArgs args) {
+
+// Survivor handling for k_scan5 (rare path; noinline keeps its registers
+// out of the hot loop's budget).
+__device__ __attribute__((noinline)) void scan5_handle_survivor(
+    const ScanArgs& args, DevCtl* ctl, const u64* s_pool, const u16* abc,
+    i64 rank, int d, int e) {
+  ttable tt5[5];
+  const int ids[5] = {abc[0], abc[1], abc[2], d, e};
+  for (int j = 0; j < 5; j++) {
+    for (int w = 0; w < 4; w++) tt5[j].w[w] = s_pool[ids[j] * PSTR + w];
+  }
+  u32 p1, p0;
+  if (!lut5_p_masks(tt5, args.T1, args.T0, &p1, &p0)) return;
+  u8 fo, fi;
+  int split;
+  if (!lut5_solve_from_p(p1, p0, dev_rnd(args.seed, rank), &fo, &fi, &split)) {
+    return;
+  }
+  if (args.count_all) return;
+  const u16 nums[5] = {abc[0], abc[1], abc[2], static_cast<u16>(d),
+                       static_cast<u16>(e)};
+  const u8* sp = SPLITS5[split];
+  u16 res[10] = {};
+  res[0] = fo;
+  res[1] = fi;
+  for (int j = 0; j < 3; j++) res[2 + j] = nums[sp[j]];
+  res[5] = nums[sp[3]];
+  res[6] = nums[sp[4]];
+  dev_publish(ctl, res);
+}
+
+__global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan5(ScanArgs args) {
   // Triple batching: one dequeue + one cooperative prefix-cell build + one
   // barrier per TB triples (measured at TB=1: 48% of wave time parked on
   // the per-triple barriers; batching amortizes them 8x).
@@ -461,89 +492,95 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan5(ScanArgs args) {
     }
     __syncthreads();
 
+    // Per-thread runs of K consecutive pairs: one decode per run, cheap
+    // lexicographic increments inside (a decode costs ~40 ops, the
+    // increment ~4). Runs rarely cross triple boundaries (re-locate
+    // handles it).
+    constexpr int K = 4;
+    const i64 nruns = (total_pairs + K - 1) / K;
     int it = 0;
-    for (i64 idx = threadIdx.x; idx < total_pairs; idx += blockDim.x) {
-      if (((it++) & 31) == 0 && !args.count_all && dev_abort(ctl)) break;
-      // Locate the triple for this flat index (TB-entry scan).
+    for (i64 run = threadIdx.x; run < nruns; run += blockDim.x) {
+      if (((it++) & 15) == 0 && !args.count_all && dev_abort(ctl)) break;
+      i64 idx = run * K;
+      const i64 idx_end = idx + K < total_pairs ? idx + K : total_pairs;
       int t = 0;
       while (s_prefix[t + 1] <= idx) t++;
-      const i64 q = s_lo[t] + (idx - s_prefix[t]);
-      const int m = s_m[t];
-      const int c = s_c[t];
+      int m = s_m[t];
+      int cgate = s_c[t];
+      i64 q = s_lo[t] + (idx - s_prefix[t]);
       int d2, e2;
       dev_decode_pair(q, m, &d2, &e2);
-      int d = c + 1 + d2, e = c + 1 + e2;
-      if (args.excl != 0) {
-        if ((d < 64 && ((args.excl >> d) & 1)) || (e < 64 && ((args.excl >> e) & 1))) {
-          continue;
-        }
-      }
-      local_eval++;
 
-      const u64* td = &s_pool[d * PSTR];
-      const u64* te = &s_pool[e * PSTR];
-      u64 td_[4], te_[4], ntd_[4], nte_[4];
-#pragma unroll
-      for (int w = 0; w < 4; w++) {
-        td_[w] = td[w];
-        te_[w] = te[w];
-        ntd_[w] = ~td_[w];
-        nte_[w] = ~te_[w];
-      }
-
-      u32 p1 = 0, p0 = 0;
-      bool ok = true;
-#pragma unroll
-      for (int u = 0; u < 8; u++) {
-        u64 r11_1 = 0, r10_1 = 0, r01_1 = 0, r00_1 = 0;
-        u64 r11_0 = 0, r10_0 = 0, r01_0 = 0, r00_0 = 0;
-#pragma unroll
-        for (int w = 0; w < 4; w++) {
-          const u64 h1 = s_H1[t][u][w];
-          const u64 h0 = s_H0[t][u][w];
-          const u64 a1 = h1 & td_[w];
-          const u64 na1 = h1 & ntd_[w];
-          const u64 a0 = h0 & td_[w];
-          const u64 na0 = h0 & ntd_[w];
-          r11_1 |= a1 & te_[w];
-          r10_1 |= a1 & nte_[w];
-          r01_1 |= na1 & te_[w];
-          r00_1 |= na1 & nte_[w];
-          r11_0 |= a0 & te_[w];
-          r10_0 |= a0 & nte_[w];
-          r01_0 |= na0 & te_[w];
-          r00_0 |= na0 & nte_[w];
+      for (; idx < idx_end; idx++) {
+        if (idx >= s_prefix[t + 1]) {
+          // Crossed into the next non-empty triple.
+          do { t++; } while (s_prefix[t + 1] <= idx);
+          m = s_m[t];
+          cgate = s_c[t];
+          q = s_lo[t] + (idx - s_prefix[t]);
+          dev_decode_pair(q, m, &d2, &e2);
         }
-        if ((r11_1 && r11_0) || (r10_1 && r10_0) || (r01_1 && r01_0) ||
-            (r00_1 && r00_0)) {
-          ok = false;
-          break;
+        const int d = cgate + 1 + d2, e = cgate + 1 + e2;
+        bool skip = false;
+        if (args.excl != 0) {
+          skip = (d < 64 && ((args.excl >> d) & 1)) ||
+                 (e < 64 && ((args.excl >> e) & 1));
         }
-        const u32 cbase = static_cast<u32>(u) << 2;
-        const u32 b1 = (r11_1 != 0 ? 8u : 0u) | (r10_1 != 0 ? 4u : 0u) |
-                       (r01_1 != 0 ? 2u : 0u) | (r00_1 != 0 ? 1u : 0u);
-        const u32 b0 = (r11_0 != 0 ? 8u : 0u) | (r10_0 != 0 ? 4u : 0u) |
-                       (r01_0 != 0 ? 2u : 0u) | (r00_0 != 0 ? 1u : 0u);
-        p1 |= b1 << cbase;
-        p0 |= b0 << cbase;
-      }
-      if (!ok) continue;
+        if (!skip) {
+          local_eval++;
+          const u64* td = &s_pool[d * PSTR];
+          const u64* te = &s_pool[e * PSTR];
+          u64 td_[4], te_[4];
+#pragma unroll
+          for (int w = 0; w < 4; w++) {
+            td_[w] = td[w];
+            te_[w] = te[w];
+          }
 
-      u8 fo, fi;
-      int split;
-      if (lut5_solve_from_p(p1, p0, dev_rnd(args.seed, s_base[t] + q), &fo, &fi,
-                            &split)) {
-        if (!args.count_all) {
-          const u16 nums[5] = {s_abc[t][0], s_abc[t][1], s_abc[t][2],
-                               static_cast<u16>(d), static_cast<u16>(e)};
-          const u8* sp = SPLITS5[split];
-          u16 res[10] = {};
-          res[0] = fo;
-          res[1] = fi;
-          for (int j = 0; j < 3; j++) res[2 + j] = nums[sp[j]];
-          res[5] = nums[sp[3]];
-          res[6] = nums[sp[4]];
-          dev_publish(ctl, res);
+          bool ok = true;
+#pragma unroll
+          for (int u = 0; u < 8; u++) {
+            u64 r11_1 = 0, r10_1 = 0, r01_1 = 0, r00_1 = 0;
+            u64 r11_0 = 0, r10_0 = 0, r01_0 = 0, r00_0 = 0;
+#pragma unroll
+            for (int w = 0; w < 4; w++) {
+              const u64 h1 = s_H1[t][u][w];
+              const u64 h0 = s_H0[t][u][w];
+              // x & ~y == x ^ (x & y): avoids materializing ~td/~te.
+              const u64 a1 = h1 & td_[w];
+              const u64 na1 = h1 ^ a1;
+              const u64 a0 = h0 & td_[w];
+              const u64 na0 = h0 ^ a0;
+              const u64 x11_1 = a1 & te_[w];
+              const u64 x01_1 = na1 & te_[w];
+              const u64 x11_0 = a0 & te_[w];
+              const u64 x01_0 = na0 & te_[w];
+              r11_1 |= x11_1;
+              r10_1 |= a1 ^ x11_1;
+              r01_1 |= x01_1;
+              r00_1 |= na1 ^ x01_1;
+              r11_0 |= x11_0;
+              r10_0 |= a0 ^ x11_0;
+              r01_0 |= x01_0;
+              r00_0 |= na0 ^ x01_0;
+            }
+            if ((r11_1 && r11_0) || (r10_1 && r10_0) || (r01_1 && r01_0) ||
+                (r00_1 && r00_0)) {
+              ok = false;
+              break;
+            }
+          }
+          if (ok) {
+            scan5_handle_survivor(args, ctl, s_pool, s_abc[t], s_base[t] + q,
+                                  d, e);
+          }
+        }
+        // Next pair in lexicographic order.
+        q++;
+        e2++;
+        if (e2 >= m) {
+          d2++;
+          e2 = d2 + 1;
         }
       }
     }
