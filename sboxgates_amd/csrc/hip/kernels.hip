@@ -598,18 +598,50 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan5(ScanArgs args) {
 // frontier resident; overflow of the per-chunk buffer is reported and the
 // host re-scans a smaller range).
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(SCAN_BLOCK) k_scan7_filter(ScanArgs args) {
-  __shared__ alignas(16) u64 s_pool[MAX_GATES * 4];
-  __shared__ alignas(16) u64 s_H1[16][4];
-  __shared__ alignas(16) u64 s_H0[16][4];
-  __shared__ i64 s_info[4];
-  __shared__ int s_abcd[4];
+// Survivor handling for k_scan7_filter: exact p-mask rebuild + hit append.
+__device__ __attribute__((noinline)) void scan7_handle_survivor(
+    const ScanArgs& args, DevCtl* ctl, const u64* s_pool, const u16* abcd,
+    int e, int f, int g) {
+  ttable tt7[7];
+  const int ids[7] = {abcd[0], abcd[1], abcd[2], abcd[3], e, f, g};
+  for (int j = 0; j < 7; j++) {
+    for (int w = 0; w < 4; w++) tt7[j].w[w] = s_pool[ids[j] * PSTR + w];
+  }
+  u64 p1[2], p0[2];
+  if (!lut7_p_masks(tt7, args.T1, args.T0, p1, p0)) return;
+  if (args.count_all) return;  // bench mode: no hit materialization
+  unsigned long long slot = __hip_atomic_fetch_add(&ctl->hit_count, 1ULL,
+                                                   __ATOMIC_RELAXED,
+                                                   __HIP_MEMORY_SCOPE_AGENT);
+  if (slot >= args.hit_cap) {
+    __hip_atomic_store(&ctl->overflow, 1u, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
+    return;
+  }
+  Hit7& h = args.hits[slot];
+  h.p1[0] = p1[0];
+  h.p1[1] = p1[1];
+  h.p0[0] = p0[0];
+  h.p0[1] = p0[1];
+  for (int j = 0; j < 7; j++) h.nums[j] = static_cast<u16>(ids[j]);
+}
+
+__global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan7_filter(ScanArgs args) {
+  // Quad batching: one dequeue + one cooperative 16-cell build per QB
+  // quadruples (same barrier-amortization pattern as k_scan5).
+  constexpr int QB = 4;
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * PSTR];
+  __shared__ alignas(16) u64 s_H1[QB][16][4];
+  __shared__ alignas(16) u64 s_H0[QB][16][4];
+  __shared__ i64 s_lo[QB], s_prefix[QB + 1];
+  __shared__ int s_m[QB], s_d[QB];
+  __shared__ u16 s_abcd[QB][4];
+  __shared__ i64 s_batch[2];  // [0]=batch base quad index, [1]=first base
+  __shared__ int s_stop;
   __shared__ unsigned long long s_eval;
 
   const int n = args.n;
-  for (int i = threadIdx.x; i < n * 4; i += blockDim.x) {
-    s_pool[i] = reinterpret_cast<const u64*>(args.pool)[i];
-  }
+  load_pool_lds(s_pool, args.pool, n);
   if (threadIdx.x == 0) s_eval = 0;
 
   DevCtl* ctl = args.ctl;
@@ -619,94 +651,117 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan7_filter(ScanArgs args) {
   for (;;) {
     __syncthreads();
     if (threadIdx.x == 0) {
-      unsigned long long t = __hip_atomic_fetch_add(&ctl->queue, 1ULL,
-                                                    __ATOMIC_RELAXED,
-                                                    __HIP_MEMORY_SCOPE_AGENT);
-      s_info[0] = -1;
-      i64 qidx = static_cast<i64>(t);
-      if (qidx < total4 &&
-          __hip_atomic_load(&ctl->overflow, __ATOMIC_RELAXED,
-                            __HIP_MEMORY_SCOPE_AGENT) == 0) {
-        // Decode quad (a,b,c,d).
-        int a = first_of_rank<cf4>(qidx, n, total4);
+      bool halt = (!args.count_all && dev_abort(ctl)) ||
+                  __hip_atomic_load(&ctl->overflow, __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT) != 0;
+      s_stop = halt ? 1 : 0;
+      if (!halt) {
+        s_batch[0] = static_cast<i64>(__hip_atomic_fetch_add(
+            &ctl->queue, static_cast<unsigned long long>(QB), __ATOMIC_RELAXED,
+            __HIP_MEMORY_SCOPE_AGENT));
+      }
+    }
+    __syncthreads();
+    if (s_stop) break;
+    const i64 batch_base = s_batch[0];
+    __syncthreads();
+
+    if (threadIdx.x < QB) {
+      const int t = threadIdx.x;
+      const i64 qidx = batch_base + t;
+      i64 base = -1, lo = 0, hi = 0;
+      int m = 0, a = 0, b = 0, c = 0, d = 0;
+      if (qidx < total4) {
+        a = first_of_rank<cf4>(qidx, n, total4);
         i64 rem = qidx - (total4 - cf4(n - a));
         int m1 = n - a - 1;
         int b0 = first_of_rank<cf3>(rem, m1, cf3(m1));
         i64 rem2 = rem - (cf3(m1) - cf3(m1 - b0));
         int c2, d2;
         dev_decode_pair(rem2, m1 - b0 - 1, &c2, &d2);
-        int b = a + 1 + b0;
-        int c = b + 1 + c2;
-        int d = b + 1 + d2;
-        // Base rank of (a,b,c,d,d+1,d+2,d+3) in C(n,7): hockey-stick sums
-        // of the skipped blocks at each level.
-        i64 base = (cf7(n) - cf7(n - a)) + (cf6(n - a - 1) - cf6(n - b)) +
-                   (cf5(n - b - 1) - cf5(n - c)) + (cf4(n - c - 1) - cf4(n - d));
-        int m = n - 1 - d;
+        b = a + 1 + b0;
+        c = b + 1 + c2;
+        d = b + 1 + d2;
+        base = (cf7(n) - cf7(n - a)) + (cf6(n - a - 1) - cf6(n - b)) +
+               (cf5(n - b - 1) - cf5(n - c)) + (cf4(n - c - 1) - cf4(n - d));
+        m = n - 1 - d;
         i64 ntrips = cf3(m);
-        i64 lo = args.begin > base ? args.begin - base : 0;
-        i64 hi = args.end - base < ntrips ? args.end - base : ntrips;
+        lo = args.begin > base ? args.begin - base : 0;
+        hi = args.end - base < ntrips ? args.end - base : ntrips;
         bool excl_prefix =
             args.excl != 0 &&
             (((a < 64) && ((args.excl >> a) & 1)) ||
              ((b < 64) && ((args.excl >> b) & 1)) ||
              ((c < 64) && ((args.excl >> c) & 1)) ||
              ((d < 64) && ((args.excl >> d) & 1)));
-        if (base >= args.end) {
-          s_info[0] = -1;
-        } else if (lo >= hi || excl_prefix) {
-          s_info[0] = -2;
-        } else {
-          s_info[0] = base;
-          s_info[1] = lo;
-          s_info[2] = hi;
-          s_info[3] = m;
-          s_abcd[0] = a;
-          s_abcd[1] = b;
-          s_abcd[2] = c;
-          s_abcd[3] = d;
-        }
+        if (base >= args.end || lo >= hi || excl_prefix) lo = hi = 0;
       }
+      if (t == 0) {
+        s_batch[1] = base;
+        s_prefix[0] = 0;
+      }
+      s_lo[t] = lo;
+      s_m[t] = m;
+      s_d[t] = d;
+      s_abcd[t][0] = static_cast<u16>(a);
+      s_abcd[t][1] = static_cast<u16>(b);
+      s_abcd[t][2] = static_cast<u16>(c);
+      s_abcd[t][3] = static_cast<u16>(d);
+      s_prefix[t + 1] = hi - lo;
     }
     __syncthreads();
-    if (s_info[0] == -1) break;
-    if (s_info[0] == -2) continue;
+    if (threadIdx.x == 0) {
+      s_stop = (batch_base >= total4 || s_batch[1] >= args.end) ? 1 : 0;
+      for (int t = 0; t < QB; t++) s_prefix[t + 1] += s_prefix[t];
+    }
+    __syncthreads();
+    if (s_stop) break;
+    const i64 total_trips = s_prefix[QB];
+    if (total_trips == 0) continue;
 
-    const int a = s_abcd[0], b = s_abcd[1], c = s_abcd[2], d = s_abcd[3];
-    const int m = static_cast<int>(s_info[3]);
-    const i64 base = s_info[0];
-    const i64 lo = s_info[1], hi = s_info[2];
-
-    // 16 prefix cells x 4 words x {1,0} = 128 items.
-    if (threadIdx.x < 128) {
-      int u = threadIdx.x & 15;
-      int w = (threadIdx.x >> 4) & 3;
-      bool is1 = threadIdx.x >= 64;
-      u64 ca = (u & 8) ? s_pool[a * 4 + w] : ~s_pool[a * 4 + w];
-      u64 cb = (u & 4) ? s_pool[b * 4 + w] : ~s_pool[b * 4 + w];
-      u64 cc = (u & 2) ? s_pool[c * 4 + w] : ~s_pool[c * 4 + w];
-      u64 cd = (u & 1) ? s_pool[d * 4 + w] : ~s_pool[d * 4 + w];
-      u64 cell = ca & cb & cc & cd;
-      if (is1) {
-        s_H1[u][w] = cell & args.T1.w[w];
-      } else {
-        s_H0[u][w] = cell & args.T0.w[w];
+    // Cooperative prefix-cell build: QB x 16 cells x 4 words x {1,0} = 512.
+    {
+      const int half = QB * 16 * 4;  // 256
+      for (int item = threadIdx.x; item < 2 * half; item += blockDim.x) {
+        const bool is1 = item < half;
+        const int iv = is1 ? item : item - half;
+        const int t = iv >> 6;
+        const int u = (iv >> 2) & 15;
+        const int w = iv & 3;
+        if (s_prefix[t + 1] == s_prefix[t]) continue;
+        const int a = s_abcd[t][0], b = s_abcd[t][1], c = s_abcd[t][2],
+                  d = s_abcd[t][3];
+        u64 ca = (u & 8) ? s_pool[a * PSTR + w] : ~s_pool[a * PSTR + w];
+        u64 cb = (u & 4) ? s_pool[b * PSTR + w] : ~s_pool[b * PSTR + w];
+        u64 cc = (u & 2) ? s_pool[c * PSTR + w] : ~s_pool[c * PSTR + w];
+        u64 cd = (u & 1) ? s_pool[d * PSTR + w] : ~s_pool[d * PSTR + w];
+        u64 cell = ca & cb & cc & cd;
+        if (is1) {
+          s_H1[t][u][w] = cell & args.T1.w[w];
+        } else {
+          s_H0[t][u][w] = cell & args.T0.w[w];
+        }
       }
     }
     __syncthreads();
 
     int it = 0;
-    for (i64 q = lo + threadIdx.x; q < hi; q += blockDim.x) {
-      if (((it++) & 31) == 0) {
+    for (i64 idx = threadIdx.x; idx < total_trips; idx += blockDim.x) {
+      if (((it++) & 15) == 0) {
         if ((!args.count_all && dev_abort(ctl)) ||
             __hip_atomic_load(&ctl->overflow, __ATOMIC_RELAXED,
                               __HIP_MEMORY_SCOPE_AGENT) != 0) {
           break;
         }
       }
+      int t = 0;
+      while (s_prefix[t + 1] <= idx) t++;
+      const i64 q = s_lo[t] + (idx - s_prefix[t]);
+      const int m = s_m[t];
+      const int dgate = s_d[t];
       int e2, f2, g2;
       dev_decode_triple(q, m, &e2, &f2, &g2);
-      int e = d + 1 + e2, f = d + 1 + f2, g = d + 1 + g2;
+      int e = dgate + 1 + e2, f = dgate + 1 + f2, g = dgate + 1 + g2;
       if (args.excl != 0) {
         if ((e < 64 && ((args.excl >> e) & 1)) || (f < 64 && ((args.excl >> f) & 1)) ||
             (g < 64 && ((args.excl >> g) & 1))) {
@@ -715,85 +770,54 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan7_filter(ScanArgs args) {
       }
       local_eval++;
 
-      u64 te_[4], tf_[4], tg_[4], nte_[4], ntf_[4], ntg_[4];
+      u64 te_[4], tf_[4], tg_[4];
 #pragma unroll
       for (int w = 0; w < 4; w++) {
-        te_[w] = s_pool[e * 4 + w];
-        tf_[w] = s_pool[f * 4 + w];
-        tg_[w] = s_pool[g * 4 + w];
-        nte_[w] = ~te_[w];
-        ntf_[w] = ~tf_[w];
-        ntg_[w] = ~tg_[w];
+        te_[w] = s_pool[e * PSTR + w];
+        tf_[w] = s_pool[f * PSTR + w];
+        tg_[w] = s_pool[g * PSTR + w];
       }
 
-      u64 p1[2] = {0, 0}, p0[2] = {0, 0};
       bool ok = true;
       for (int u = 0; u < 16 && ok; u++) {
-        // 8 sub-patterns of (e, f, g) per prefix cell.
         u64 acc1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
         u64 acc0[8] = {0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
         for (int w = 0; w < 4; w++) {
-          const u64 h1 = s_H1[u][w];
-          const u64 h0 = s_H0[u][w];
-          const u64 e1 = h1 & te_[w], e0n = h1 & nte_[w];
-          const u64 z1 = h0 & te_[w], z0n = h0 & nte_[w];
-          const u64 ef11 = e1 & tf_[w], ef10 = e1 & ntf_[w];
-          const u64 ef01 = e0n & tf_[w], ef00 = e0n & ntf_[w];
-          const u64 zf11 = z1 & tf_[w], zf10 = z1 & ntf_[w];
-          const u64 zf01 = z0n & tf_[w], zf00 = z0n & ntf_[w];
-          acc1[7] |= ef11 & tg_[w];
-          acc1[6] |= ef11 & ntg_[w];
-          acc1[5] |= ef10 & tg_[w];
-          acc1[4] |= ef10 & ntg_[w];
-          acc1[3] |= ef01 & tg_[w];
-          acc1[2] |= ef01 & ntg_[w];
-          acc1[1] |= ef00 & tg_[w];
-          acc1[0] |= ef00 & ntg_[w];
-          acc0[7] |= zf11 & tg_[w];
-          acc0[6] |= zf11 & ntg_[w];
-          acc0[5] |= zf10 & tg_[w];
-          acc0[4] |= zf10 & ntg_[w];
-          acc0[3] |= zf01 & tg_[w];
-          acc0[2] |= zf01 & ntg_[w];
-          acc0[1] |= zf00 & tg_[w];
-          acc0[0] |= zf00 & ntg_[w];
+          const u64 h1 = s_H1[t][u][w];
+          const u64 h0 = s_H0[t][u][w];
+          const u64 e1 = h1 & te_[w];
+          const u64 e0n = h1 ^ e1;
+          const u64 z1 = h0 & te_[w];
+          const u64 z0n = h0 ^ z1;
+          const u64 ef11 = e1 & tf_[w];
+          const u64 ef10 = e1 ^ ef11;
+          const u64 ef01 = e0n & tf_[w];
+          const u64 ef00 = e0n ^ ef01;
+          const u64 zf11 = z1 & tf_[w];
+          const u64 zf10 = z1 ^ zf11;
+          const u64 zf01 = z0n & tf_[w];
+          const u64 zf00 = z0n ^ zf01;
+          u64 x;
+          x = ef11 & tg_[w]; acc1[7] |= x; acc1[6] |= ef11 ^ x;
+          x = ef10 & tg_[w]; acc1[5] |= x; acc1[4] |= ef10 ^ x;
+          x = ef01 & tg_[w]; acc1[3] |= x; acc1[2] |= ef01 ^ x;
+          x = ef00 & tg_[w]; acc1[1] |= x; acc1[0] |= ef00 ^ x;
+          x = zf11 & tg_[w]; acc0[7] |= x; acc0[6] |= zf11 ^ x;
+          x = zf10 & tg_[w]; acc0[5] |= x; acc0[4] |= zf10 ^ x;
+          x = zf01 & tg_[w]; acc0[3] |= x; acc0[2] |= zf01 ^ x;
+          x = zf00 & tg_[w]; acc0[1] |= x; acc0[0] |= zf00 ^ x;
         }
 #pragma unroll
-        for (int p = 0; p < 8; p++) {
-          bool h1 = acc1[p] != 0, h0 = acc0[p] != 0;
-          if (h1 && h0) {
+        for (int pp = 0; pp < 8; pp++) {
+          if (acc1[pp] && acc0[pp]) {
             ok = false;
             break;
           }
-          int cell = (u << 3) | p;
-          if (h1) p1[cell >> 6] |= 1ULL << (cell & 63);
-          if (h0) p0[cell >> 6] |= 1ULL << (cell & 63);
         }
       }
       if (!ok) continue;
-
-      // Survivor: append to hit buffer.
-      unsigned long long slot = __hip_atomic_fetch_add(&ctl->hit_count, 1ULL,
-                                                       __ATOMIC_RELAXED,
-                                                       __HIP_MEMORY_SCOPE_AGENT);
-      if (slot >= args.hit_cap) {
-        __hip_atomic_store(&ctl->overflow, 1u, __ATOMIC_RELAXED,
-                           __HIP_MEMORY_SCOPE_AGENT);
-      } else {
-        Hit7& h = args.hits[slot];
-        h.p1[0] = p1[0];
-        h.p1[1] = p1[1];
-        h.p0[0] = p0[0];
-        h.p0[1] = p0[1];
-        h.nums[0] = static_cast<u16>(a);
-        h.nums[1] = static_cast<u16>(b);
-        h.nums[2] = static_cast<u16>(c);
-        h.nums[3] = static_cast<u16>(d);
-        h.nums[4] = static_cast<u16>(e);
-        h.nums[5] = static_cast<u16>(f);
-        h.nums[6] = static_cast<u16>(g);
-      }
+      scan7_handle_survivor(args, ctl, s_pool, s_abcd[t], e, f, g);
     }
   }
 

@@ -68,8 +68,8 @@ ScanResult cpu_scan5(const ScanRequest& rq, i64 begin, i64 end) {
   for (int j = 0; j < 5; j++) tt[j] = rq.tables[nums[j]];
 
   for (i64 i = begin; i < end; i++) {
-    out.evaluated++;
     if (!excluded(rq, nums, 5)) {
+      out.evaluated++;
       u32 p1, p0;
       if (lut5_p_masks(tt, T1, T0, &p1, &p0)) {
         u8 fo, fi;
@@ -108,8 +108,8 @@ ScanResult cpu_scan7(const ScanRequest& rq, i64 begin, i64 end) {
   for (int j = 0; j < 7; j++) tt[j] = rq.tables[nums[j]];
 
   for (i64 i = begin; i < end; i++) {
-    out.evaluated++;
     if (!excluded(rq, nums, 7)) {
+      out.evaluated++;
       u64 p1[2], p0[2];
       if (lut7_p_masks(tt, T1, T0, p1, p0)) {
         u64 rnd = hash_mix64(rq.seed ^ static_cast<u64>(i));

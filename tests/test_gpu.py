@@ -226,3 +226,49 @@ def test_rijndael_gate_mode_bit0_gpu():
     st.set_output(0, out)
     assert validate_circuit(st, sbox, n, bit=0)
     assert eng.stats()["gpu_scans"] > 0
+
+
+def test_scan7_hit_buffer_overflow_splitting(engines, monkeypatch):
+    """A tiny hit cap must trigger range splitting, not truncation: the
+    planted solution is still found."""
+    import os
+    import subprocess
+    import sys
+    # The cap is read at GpuEngine creation; run in a subprocess with the
+    # env var set.
+    code = r'''
+import sys
+sys.path.insert(0, ".")
+import random
+from sboxgates_amd import models
+from sboxgates_amd.ops import (gen_lut_ttable, make_engine, mask_for_inputs,
+                               n_choose_k, tt_eq_mask)
+eng = make_engine(lut_graph=True, seed=1, gpu="force", save_states=False)
+sbox, n = models.load("rijndael")
+eng.set_sbox(sbox, n)
+st = eng.initial_state()
+st.grow_pool_random(20, 77)
+mask = mask_for_inputs(8)
+rng = random.Random(9)
+ids = sorted(rng.sample(range(20), 7))
+tabs = [st.gate(i)["table"] for i in ids]
+t_o = gen_lut_ttable(0x3C, tabs[0], tabs[1], tabs[2])
+t_m = gen_lut_ttable(0xA5, tabs[3], tabs[4], tabs[5])
+target = gen_lut_ttable(0x96, t_o, t_m, tabs[6])
+found, res, ev = eng.scan_pool(7, st, target, mask, 0, n_choose_k(20, 7), seed=3)
+assert found, "planted 7-LUT not found under tiny hit cap"
+g_o = gen_lut_ttable(res[0], st.gate(res[3])["table"], st.gate(res[4])["table"],
+                     st.gate(res[5])["table"])
+g_m = gen_lut_ttable(res[1], st.gate(res[6])["table"], st.gate(res[7])["table"],
+                     st.gate(res[8])["table"])
+g_i = gen_lut_ttable(res[2], g_o, g_m, st.gate(res[9])["table"])
+assert tt_eq_mask(target, g_i, mask)
+print("overflow-split ok")
+'''
+    env = dict(os.environ)
+    env["SBOXGATES_HIT_CAP"] = "64"
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300,
+                       cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr
+    assert "overflow-split ok" in r.stdout
