@@ -24,7 +24,8 @@ def main():
     ctx = make_ctx()
 
     sbox, n = models.load("des_s1")
-    eng = make_engine(lut_graph=True, seed=21, gpu="off", save_states=False,
+    gpu_mode = "force" if os.environ.get("SBOXGATES_TEST_GPU") else "off"
+    eng = make_engine(lut_graph=True, seed=21, gpu=gpu_mode, save_states=False,
                       oneoutput=0, ctx=ctx)
     eng.set_sbox(sbox, n)
 

@@ -272,3 +272,38 @@ print("overflow-split ok")
                        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     assert r.returncode == 0, r.stderr
     assert "overflow-split ok" in r.stdout
+
+
+def test_distributed_gpu_search_two_ranks():
+    """World-2 distributed LUT search with BOTH ranks driving GPU kernels
+    (gloo control plane, both ranks on device 0 — the protocol and kernel
+    interaction exactly match the multi-GPU RCCL deployment)."""
+    import json
+    import os
+    import subprocess
+    import sys
+    import tempfile
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    worker = os.path.join(repo, "tests", "_dist_worker.py")
+    with tempfile.TemporaryDirectory() as d:
+        out = os.path.join(d, "result")
+        procs = []
+        for rank in range(2):
+            env = dict(os.environ)
+            env.update({
+                "RANK": str(rank), "WORLD_SIZE": "2", "LOCAL_RANK": str(rank),
+                "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29871",
+                "SBOXGATES_TEST_GPU": "1",
+                "SBOXGATES_CHUNK5": "20000", "SBOXGATES_CHUNK7": "100000",
+            })
+            procs.append(subprocess.Popen(
+                [sys.executable, worker, "one_output_search", out], env=env,
+                cwd=d, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                text=True))
+        for rank, p in enumerate(procs):
+            stdout, stderr = p.communicate(timeout=300)
+            assert p.returncode == 0, f"rank {rank}:\n{stderr}"
+        with open(out + ".rank0") as f:
+            r0 = json.load(f)
+        assert r0["ok"]
+        assert r0["stats"]["gpu_scans"] > 0, "GPU kernels must be used"
