@@ -990,10 +990,17 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
     gatenum first[5];
     nth_combination(begin, rq.n, 5, 0, first);
     i64 t_begin = combination_rank(first, 3, rq.n);
+    gatenum last[5];
+    nth_combination(end - 1, rq.n, 5, 0, last);
+    i64 t_last = combination_rank(last, 3, rq.n);
     im->h_ctl->queue = static_cast<unsigned long long>(t_begin);
     SBG_HIP_CHECK(hipMemcpyAsync(im->d_ctl, im->h_ctl, sizeof(DevCtl),
                                  hipMemcpyHostToDevice, im->stream));
-    int grid = 2048;
+    // Size the grid to the triple count: small scans (deep recursion) pay
+    // for pool staging per block, so an always-2048 grid costs ~10x the
+    // useful work there.
+    i64 tcount = t_last - t_begin + 1;
+    int grid = static_cast<int>(std::min<i64>((tcount + 7) / 8 + 1, 2048));
     hipLaunchKernelGGL(k_scan5, dim3(grid), dim3(SCAN_BLOCK), 0, im->stream, args);
     SBG_HIP_CHECK(hipGetLastError());
     SBG_HIP_CHECK(hipMemcpyAsync(im->h_ctl, im->d_ctl, sizeof(DevCtl),
@@ -1008,14 +1015,19 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
         std::memset(im->h_ctl, 0, sizeof(DevCtl));
         gatenum first[7];
         nth_combination(lo, rq.n, 7, 0, first);
-        im->h_ctl->queue =
-            static_cast<unsigned long long>(combination_rank(first, 4, rq.n));
+        i64 q_begin = combination_rank(first, 4, rq.n);
+        gatenum last[7];
+        nth_combination(hi - 1, rq.n, 7, 0, last);
+        i64 q_last = combination_rank(last, 4, rq.n);
+        im->h_ctl->queue = static_cast<unsigned long long>(q_begin);
         SBG_HIP_CHECK(hipMemcpyAsync(im->d_ctl, im->h_ctl, sizeof(DevCtl),
                                      hipMemcpyHostToDevice, im->stream));
         ScanArgs a2 = args;
         a2.begin = lo;
         a2.end = hi;
-        hipLaunchKernelGGL(k_scan7_filter, dim3(2048), dim3(SCAN_BLOCK), 0,
+        int grid7 = static_cast<int>(
+            std::min<i64>((q_last - q_begin + 1 + 3) / 4 + 1, 2048));
+        hipLaunchKernelGGL(k_scan7_filter, dim3(grid7), dim3(SCAN_BLOCK), 0,
                            im->stream, a2);
         SBG_HIP_CHECK(hipGetLastError());
         SBG_HIP_CHECK(hipMemcpyAsync(im->h_ctl, im->d_ctl, sizeof(DevCtl),

@@ -254,9 +254,14 @@ void Engine::set_sbox(const u8 sbox[256], int num_inputs) {
 }
 
 ScanResult Engine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
-  constexpr i64 GPU_MIN_CANDIDATES = 1 << 16;
-  bool use_gpu = gpu_ != nullptr &&
-                 (opt_.gpu == GPU_FORCE || end - begin >= GPU_MIN_CANDIDATES);
+  // Per-k GPU cutover: per-combination cost grows steeply with k (a 7-LUT
+  // feasibility check walks 128 cells vs 32 for 5-LUT and 8 for 3-LUT), so
+  // the range size where the GPU (incl. ~20-40us launch+upload overhead)
+  // beats the CPU differs by ~100x across kinds. Measured on the AES bit-0
+  // search: a size-only 2^16 threshold left 2.1e9 7-LUT combos on the CPU
+  // and tripled the wall time.
+  const i64 gpu_min = k == 7 ? 256 : (k == 5 ? 4096 : 1 << 14);
+  bool use_gpu = gpu_ != nullptr && (opt_.gpu == GPU_FORCE || end - begin >= gpu_min);
   const auto t0 = std::chrono::steady_clock::now();
   ScanResult r;
   if (use_gpu) {
