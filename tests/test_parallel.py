@@ -56,7 +56,7 @@ def test_distributed_lut_search(tmp_path):
 def test_distributed_multi_chunk(tmp_path):
     """Tiny chunks force many allreduce rounds through the protocol."""
     res = launch("one_output_search", tmp_path,
-                 extra_env={"SBOXGATES_CHUNK5": "500", "SBOXGATES_CHUNK7": "2000"})
+                 extra_env={"SBOXGATES_CHUNK5": "2000", "SBOXGATES_CHUNK7": "20000"})
     assert res[0]["ok"] and res[1]["ok"]
 
 
