@@ -265,3 +265,13 @@ class TestScan4:
         # With only AND compositions over a random pool, realizing an AES
         # output bit exactly is essentially impossible.
         assert not found
+
+
+def test_find_circuit_api(tmp_path):
+    from sboxgates_amd.search import find_circuit
+    st = find_circuit("des_s1", bit=0, seed=3, gpu="off")
+    assert st.outputs[0] >= 0
+    st2 = find_circuit("crypto1_fa", bit=0, lut=True, seed=4, gpu="off")
+    assert st2.outputs[0] >= 0
+    st3 = find_circuit([1, 0, 2, 3], bit=1, seed=5, gpu="off")
+    assert st3.outputs[1] >= 0
