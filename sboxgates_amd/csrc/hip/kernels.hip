@@ -833,18 +833,28 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan7_filter(ScanArgs args) {
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(SCAN_BLOCK) k_scan7_assign(
     const Hit7* hits, unsigned long long nhits, DevCtl* ctl, u64 seed) {
-  const u64 nitems = nhits * LUT7_NUM_ORDERINGS;
+  // One thread per (hit, ordering, fm-octant): the 256-middle-function
+  // sweep splits across 8 threads (disjoint 32-function slices of the
+  // same shuffled order), which bounds the divergent tail of exhaustive
+  // items at ~32 colorings and feeds the GPU 8x more parallel slack.
+  constexpr u64 FM_SPLIT = 8;
+  const u64 nitems = nhits * LUT7_NUM_ORDERINGS * FM_SPLIT;
   const u64 stride = static_cast<u64>(gridDim.x) * blockDim.x;
   int tick = 0;
   for (u64 item = blockIdx.x * static_cast<u64>(blockDim.x) + threadIdx.x;
        item < nitems; item += stride) {
-    if (((tick++) & 15) == 0 && dev_abort(ctl)) return;
-    const Hit7& h = hits[item / LUT7_NUM_ORDERINGS];
-    int oidx = static_cast<int>(item % LUT7_NUM_ORDERINGS);
+    if (((tick++) & 3) == 0 && dev_abort(ctl)) return;
+    const u64 base_item = item / FM_SPLIT;
+    const int oct = static_cast<int>(item % FM_SPLIT);
+    const Hit7& h = hits[base_item / LUT7_NUM_ORDERINGS];
+    int oidx = static_cast<int>(base_item % LUT7_NUM_ORDERINGS);
     u8 ord[7];
     lut7_ordering(oidx, ord);
     u8 fo, fm, fi;
-    if (lut7_solve_ordering(h.p1, h.p0, ord, dev_rnd(seed, item), &fo, &fm, &fi)) {
+    // rnd keyed by (hit, ordering) so the 8 octants partition the same
+    // shuffled middle-function order disjointly.
+    if (lut7_solve_ordering(h.p1, h.p0, ord, dev_rnd(seed, base_item), &fo,
+                            &fm, &fi, oct * 32, 32)) {
       u16 res[10];
       res[0] = fo;
       res[1] = fm;
@@ -1044,7 +1054,7 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
       unsigned long long nhits = im->h_ctl->hit_count;
       if (nhits > im->hit_cap) nhits = im->hit_cap;
       if (nhits > 0 && !rq.count_all) {
-        u64 items = nhits * LUT7_NUM_ORDERINGS;
+        u64 items = nhits * LUT7_NUM_ORDERINGS * 8;
         int grid = static_cast<int>(
             std::min<u64>((items + SCAN_BLOCK - 1) / SCAN_BLOCK, 4096));
         hipLaunchKernelGGL(k_scan7_assign, dim3(grid), dim3(SCAN_BLOCK), 0,

@@ -270,7 +270,8 @@ constexpr int LUT7_NUM_ORDERINGS = 70;
 // fo (outer), fm (middle), fi (inner).
 SBG_HD inline bool lut7_solve_ordering(const u64 p1[2], const u64 p0[2],
                                        const u8 ord[7], u64 rnd, u8* fo_out,
-                                       u8* fm_out, u8* fi_out) {
+                                       u8* fm_out, u8* fi_out,
+                                       int fm_offset = 0, int fm_count = 256) {
   // Aggregate to (u = outer cell, w = middle cell, g): bit w of B1[u][g].
   u8 B1[8][2] = {};
   u8 B0[8][2] = {};
@@ -287,7 +288,7 @@ SBG_HD inline bool lut7_solve_ordering(const u64 p1[2], const u64 p0[2],
     if (b0) B0[u][gg] |= 1u << w;
   }
 
-  for (int fmi = 0; fmi < 256; fmi++) {
+  for (int fmi = fm_offset; fmi < fm_offset + fm_count; fmi++) {
     u8 fm = static_cast<u8>((fmi + (rnd >> 32)) & 0xff);
     // Layers for the outer coloring: (m, g) in {0,1}^2.
     // L1[layer] bit u = exists middle-cell w with fm-class m and p1 set.
