@@ -148,7 +148,8 @@ def main():
                 "combinations_per_step": total,
                 "global_batch": total,
                 "seq_len": 256,
-                "parallelism": f"dp{world} (combination-space split, RCCL)",
+                "parallelism": f"dp{world} (combination-space split over "
+                               f"{'RCCL/xGMI' if use_gpu and distributed else ('RCCL' if use_gpu else 'gloo')})",
                 "device": "gpu" if use_gpu else "cpu",
             },
         }
