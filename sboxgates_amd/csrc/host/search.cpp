@@ -3,6 +3,7 @@
 
 #include "sbg/search.hpp"
 
+#include <algorithm>
 #include <cassert>
 #include <chrono>
 #include <climits>
@@ -862,7 +863,7 @@ void Engine::generate_graph(const state& st_in) {
               num_out_states = 0;
             }
             if (st.num_gates <= best_metric_g) {
-              if (num_out_states < 20) {
+              if (num_out_states < std::min(20, opt_.beam)) {
                 out_states[num_out_states++] = st;
               } else if (opt_.verbosity >= 0) {
                 std::printf("Output state buffer full! Throwing away valid state.\n");
@@ -874,7 +875,7 @@ void Engine::generate_graph(const state& st_in) {
               num_out_states = 0;
             }
             if (st.sat_metric <= best_metric_i) {
-              if (num_out_states < 20) {
+              if (num_out_states < std::min(20, opt_.beam)) {
                 out_states[num_out_states++] = st;
               } else if (opt_.verbosity >= 0) {
                 std::printf("Output state buffer full! Throwing away valid state.\n");

@@ -42,6 +42,8 @@ struct options {
   gpu_mode_t gpu = GPU_AUTO;
   int gpu_device = -1;      // HIP device index; -1 = current device
   int num_gpus = 1;         // CLI --gpus: in-process devices (threads)
+  int beam = 20;            // --beam: tied-state beam width (<= 20; the
+                            // reference hard-codes 20, sboxgates.c:704)
   std::string output_dir;   // where XML checkpoints are written ("" = CWD)
   bool save_states = true;  // library callers may disable checkpoint writes
 
