@@ -70,14 +70,6 @@ class PyDistCtx : public DistCtx {
   py::function bcast_fn_, min_fn_;
 };
 
-boolfunc* find_fun(options& o, int which, int idx) {
-  switch (which) {
-    case 0: return &o.avail_gates[idx];
-    case 1: return &o.avail_not[idx];
-    default: return &o.avail_3[idx];
-  }
-}
-
 py::dict fun_to_dict(const boolfunc& f) {
   py::dict d;
   d["num_inputs"] = f.num_inputs;

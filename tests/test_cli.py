@@ -146,3 +146,11 @@ def test_multi_device_threads(tmp_path):
 
 def test_bad_gpus_value():
     assert run(["--gpus", "0", des()]).returncode != 0
+
+
+def test_beam_option(tmp_path):
+    r = run(["--beam", "1", "--seed", "9", "--cpu", des()], cwd=str(tmp_path),
+            timeout=300)
+    assert r.returncode == 0, r.stderr
+    assert run(["--beam", "0", des()]).returncode != 0
+    assert run(["--beam", "21", des()]).returncode != 0
