@@ -879,6 +879,9 @@ struct GpuEngine::Impl {
   DevCtl* d_ctl = nullptr;
   Hit7* d_hits = nullptr;
   Avail3Matcher* d_matcher = nullptr;
+  ttable* h_pool = nullptr;  // pinned staging: pageable-source async
+                             // copies cost ~100s of us on small scans
+  const Avail3Matcher* last_matcher = nullptr;  // uploaded-matcher cache
   u64 hit_cap = 0;
   DevCtl* h_ctl = nullptr;  // pinned staging
   std::string name;
@@ -889,6 +892,7 @@ struct GpuEngine::Impl {
     if (d_ctl != nullptr) (void)hipFree(d_ctl);
     if (d_hits != nullptr) (void)hipFree(d_hits);
     if (h_ctl != nullptr) (void)hipHostFree(h_ctl);
+    if (h_pool != nullptr) (void)hipHostFree(h_pool);
     if (stream != nullptr) (void)hipStreamDestroy(stream);
   }
 };
@@ -926,6 +930,7 @@ std::unique_ptr<GpuEngine> GpuEngine::create(int device, std::string* err) {
     SBG_HIP_CHECK(hipMalloc(&impl->d_ctl, sizeof(DevCtl)));
     SBG_HIP_CHECK(hipMalloc(&impl->d_matcher, sizeof(Avail3Matcher)));
     SBG_HIP_CHECK(hipHostMalloc(&impl->h_ctl, sizeof(DevCtl)));
+    SBG_HIP_CHECK(hipHostMalloc(&impl->h_pool, sizeof(ttable) * MAX_GATES));
     // Hit buffer for the 7-LUT frontier: default 16M hits (768 MB) per
     // chunk; overridable for memory-constrained runs.
     const char* cap_env = std::getenv("SBOXGATES_HIT_CAP");
@@ -954,8 +959,9 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   Impl* im = impl_;
   SBG_HIP_CHECK(hipSetDevice(im->device));
 
-  // Upload the pool and reset the control block.
-  SBG_HIP_CHECK(hipMemcpyAsync(im->d_pool, rq.tables, sizeof(ttable) * rq.n,
+  // Upload the pool (through pinned staging) and reset the control block.
+  std::memcpy(im->h_pool, rq.tables, sizeof(ttable) * rq.n);
+  SBG_HIP_CHECK(hipMemcpyAsync(im->d_pool, im->h_pool, sizeof(ttable) * rq.n,
                                hipMemcpyHostToDevice, im->stream));
   std::memset(im->h_ctl, 0, sizeof(DevCtl));
 
@@ -977,9 +983,12 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   if (k == 3 || k == 4) {
     if (k == 4) {
       if (rq.matcher == nullptr) throw std::runtime_error("scan4 needs matcher");
-      SBG_HIP_CHECK(hipMemcpyAsync(im->d_matcher, rq.matcher,
-                                   sizeof(Avail3Matcher), hipMemcpyHostToDevice,
-                                   im->stream));
+      if (im->last_matcher != rq.matcher) {
+        // The matcher is per-engine and immutable; upload once.
+        SBG_HIP_CHECK(hipMemcpy(im->d_matcher, rq.matcher, sizeof(Avail3Matcher),
+                                hipMemcpyHostToDevice));
+        im->last_matcher = rq.matcher;
+      }
     }
     SBG_HIP_CHECK(hipMemcpyAsync(im->d_ctl, im->h_ctl, sizeof(DevCtl),
                                  hipMemcpyHostToDevice, im->stream));
