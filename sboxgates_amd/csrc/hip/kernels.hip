@@ -218,11 +218,17 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan3(ScanArgs args) {
     int b = a + 1 + b2, c = a + 1 + c2;
 
     local_eval++;
-    const ttable* ta = reinterpret_cast<const ttable*>(&s_pool[a * PSTR]);
-    const ttable* tb = reinterpret_cast<const ttable*>(&s_pool[b * PSTR]);
-    const ttable* tc = reinterpret_cast<const ttable*>(&s_pool[c * PSTR]);
+    // Copy out of LDS: the padded 48 B stride is not ttable-aligned, so a
+    // reinterpret_cast would be UB.
+    ttable ta, tb, tc;
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      ta.w[w] = s_pool[a * PSTR + w];
+      tb.w[w] = s_pool[b * PSTR + w];
+      tc.w[w] = s_pool[c * PSTR + w];
+    }
     u32 p1, p0;
-    if (lut3_p_masks(*ta, *tb, *tc, args.T1, args.T0, &p1, &p0)) {
+    if (lut3_p_masks(ta, tb, tc, args.T1, args.T0, &p1, &p0)) {
       if (!args.count_all) {
         u8 func = lut3_function_from_p(p1, p0, dev_rnd(args.seed, idx));
         u16 res[10] = {};
@@ -299,11 +305,15 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
     int b = a + 1 + b2, c = a + 1 + c2;
 
     local_eval++;
-    const ttable* ta = reinterpret_cast<const ttable*>(&s_pool[a * PSTR]);
-    const ttable* tb = reinterpret_cast<const ttable*>(&s_pool[b * PSTR]);
-    const ttable* tc = reinterpret_cast<const ttable*>(&s_pool[c * PSTR]);
+    ttable ta, tb, tc;
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      ta.w[w] = s_pool[a * PSTR + w];
+      tb.w[w] = s_pool[b * PSTR + w];
+      tc.w[w] = s_pool[c * PSTR + w];
+    }
     u32 p1, p0;
-    if (!lut3_p_masks(*ta, *tb, *tc, args.T1, args.T0, &p1, &p0)) continue;
+    if (!lut3_p_masks(ta, tb, tc, args.T1, args.T0, &p1, &p0)) continue;
     if (args.count_all) continue;
     const u8 req1 = static_cast<u8>(p1);
     const u8 care = static_cast<u8>(p1 | p0);
