@@ -171,6 +171,7 @@ struct ScanArgs {
   u64 excl;  // excluded gate ids < 64
   u64 seed;
   int count_all;
+  int slices7;  // K7: sub-quad slicing factor (dequeue unit = quad x slice)
 };
 
 constexpr int SCAN_BLOCK = 256;
@@ -678,7 +679,10 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan7_filter(ScanArgs args) {
 
     if (threadIdx.x < QB) {
       const int t = threadIdx.x;
-      const i64 qidx = batch_base + t;
+      const int S = args.slices7;
+      const i64 unit = batch_base + t;
+      const i64 qidx = unit / S;
+      const int slice = static_cast<int>(unit % S);
       i64 base = -1, lo = 0, hi = 0;
       int m = 0, a = 0, b = 0, c = 0, d = 0;
       if (qidx < total4) {
@@ -698,6 +702,15 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan7_filter(ScanArgs args) {
         i64 ntrips = cf3(m);
         lo = args.begin > base ? args.begin - base : 0;
         hi = args.end - base < ntrips ? args.end - base : ntrips;
+        if (S > 1 && hi > lo) {
+          // This unit covers one slice of the quad's triple window.
+          const i64 span = hi - lo;
+          const i64 per = (span + S - 1) / S;
+          const i64 s_lo2 = lo + per * slice;
+          const i64 s_hi2 = s_lo2 + per < hi ? s_lo2 + per : hi;
+          lo = s_lo2 < hi ? s_lo2 : hi;
+          hi = s_hi2;
+        }
         bool excl_prefix =
             args.excl != 0 &&
             (((a < 64) && ((args.excl >> a) & 1)) ||
@@ -721,7 +734,9 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan7_filter(ScanArgs args) {
     }
     __syncthreads();
     if (threadIdx.x == 0) {
-      s_stop = (batch_base >= total4 || s_batch[1] >= args.end) ? 1 : 0;
+      s_stop = (batch_base / args.slices7 >= total4 || s_batch[1] >= args.end)
+                   ? 1
+                   : 0;
       for (int t = 0; t < QB; t++) s_prefix[t + 1] += s_prefix[t];
     }
     __syncthreads();
@@ -989,6 +1004,7 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   args.excl = rq.excl_low64;
   args.seed = rq.seed;
   args.count_all = rq.count_all ? 1 : 0;
+  args.slices7 = 1;
 
   if (k == 3 || k == 4) {
     if (k == 4) {
@@ -1048,14 +1064,22 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
         gatenum last[7];
         nth_combination(hi - 1, rq.n, 7, 0, last);
         i64 q_last = combination_rank(last, 4, rq.n);
-        im->h_ctl->queue = static_cast<unsigned long long>(q_begin);
-        SBG_HIP_CHECK(hipMemcpyAsync(im->d_ctl, im->h_ctl, sizeof(DevCtl),
-                                     hipMemcpyHostToDevice, im->stream));
+        i64 qcount = q_last - q_begin + 1;
+        // Sub-quad slicing: a single quad prefix can hold C(n-4,3) triples
+        // (tens of thousands); without slicing a small scan runs on a
+        // handful of blocks and leaves the chip idle.
+        i64 want_units = (hi - lo + 8 * SCAN_BLOCK - 1) / (8 * SCAN_BLOCK);
+        int slices = static_cast<int>(
+            std::clamp<i64>(want_units / std::max<i64>(qcount, 1), 1, 256));
         ScanArgs a2 = args;
         a2.begin = lo;
         a2.end = hi;
+        a2.slices7 = slices;
+        im->h_ctl->queue = static_cast<unsigned long long>(q_begin * slices);
+        SBG_HIP_CHECK(hipMemcpyAsync(im->d_ctl, im->h_ctl, sizeof(DevCtl),
+                                     hipMemcpyHostToDevice, im->stream));
         int grid7 = static_cast<int>(
-            std::min<i64>((q_last - q_begin + 1 + 3) / 4 + 1, 2048));
+            std::min<i64>((qcount * slices + 3) / 4 + 1, 2048));
         hipLaunchKernelGGL(k_scan7_filter, dim3(grid7), dim3(SCAN_BLOCK), 0,
                            im->stream, a2);
         SBG_HIP_CHECK(hipGetLastError());
