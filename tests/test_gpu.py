@@ -307,3 +307,24 @@ def test_distributed_gpu_search_two_ranks():
             r0 = json.load(f)
         assert r0["ok"]
         assert r0["stats"]["gpu_scans"] > 0, "GPU kernels must be used"
+
+
+def test_random_window_count_parity(engines):
+    """Counts over random [begin, end) windows must match the CPU scans
+    exactly for every k (exercises prefix/slice boundary math)."""
+    import random
+    gpu, cpu = engines
+    rng = random.Random(123)
+    st = make_pool(gpu, 60, 0xFACE)
+    target = gpu.target(0)
+    mask = mask_for_inputs(8)
+    for k in (3, 4, 5, 7):
+        total = n_choose_k(60, k if k != 4 else 3)
+        for _ in range(6):
+            a = rng.randrange(total)
+            b = min(total, a + rng.choice([1, 7, 1000, 50_000, 2_000_000]))
+            f_g, _, ev_g = gpu.scan_pool(k, st, target, mask, a, b,
+                                         count_all=True)
+            f_c, _, ev_c = cpu.scan_pool(k, st, target, mask, a, b,
+                                         count_all=True)
+            assert ev_g == ev_c == b - a, (k, a, b, ev_g, ev_c)
