@@ -24,7 +24,7 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-POOL_GATES = 300
+POOL_GATES = 450
 POOL_SEED = 0x5B0C5EED
 SCAN_SEED = 12345
 
