@@ -399,6 +399,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan5(ScanArgs args) {
   __shared__ i64 s_lo[TB], s_prefix[TB + 1];
   __shared__ int s_m[TB], s_c[TB];
   __shared__ u16 s_abc[TB][3];
+  __shared__ u32 s_live[TB];     // cells with both forced-1 and forced-0
+                                 // content; only these can contradict
   __shared__ int s_stop;         // all triples past range end
   __shared__ unsigned long long s_eval;
 
@@ -501,6 +503,18 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan5(ScanArgs args) {
         }
       }
     }
+    if (threadIdx.x < TB) s_live[threadIdx.x] = 0;
+    __syncthreads();
+    // Liveness: a cell whose H1 or H0 side is empty can never contradict
+    // (and survivors rebuild exact p-masks anyway). Sparse-mask scans in
+    // deep recursion typically keep only a few of the 8 cells live.
+    if (threadIdx.x < TB * 8) {
+      const int t = threadIdx.x >> 3;
+      const int u = threadIdx.x & 7;
+      const u64 h1 = s_H1[t][u][0] | s_H1[t][u][1] | s_H1[t][u][2] | s_H1[t][u][3];
+      const u64 h0 = s_H0[t][u][0] | s_H0[t][u][1] | s_H0[t][u][2] | s_H0[t][u][3];
+      if (h1 != 0 && h0 != 0) atomicOr(&s_live[t], 1u << u);
+    }
     __syncthreads();
 
     // Per-thread runs of K consecutive pairs: one decode per run, cheap
@@ -549,8 +563,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan5(ScanArgs args) {
           }
 
           bool ok = true;
-#pragma unroll
-          for (int u = 0; u < 8; u++) {
+          for (u32 lm = s_live[t]; lm != 0; lm &= lm - 1) {
+            const int u = __ffs(lm) - 1;
             u64 r11_1 = 0, r10_1 = 0, r01_1 = 0, r00_1 = 0;
             u64 r11_0 = 0, r10_0 = 0, r01_0 = 0, r00_0 = 0;
 #pragma unroll
@@ -647,6 +661,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan7_filter(ScanArgs args) {
   __shared__ i64 s_lo[QB], s_prefix[QB + 1];
   __shared__ int s_m[QB], s_d[QB];
   __shared__ u16 s_abcd[QB][4];
+  __shared__ u32 s_live[QB];
   __shared__ i64 s_batch[2];  // [0]=batch base quad index, [1]=first base
   __shared__ int s_stop;
   __shared__ unsigned long long s_eval;
@@ -768,6 +783,15 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan7_filter(ScanArgs args) {
         }
       }
     }
+    if (threadIdx.x < QB) s_live[threadIdx.x] = 0;
+    __syncthreads();
+    if (threadIdx.x < QB * 16) {
+      const int t = threadIdx.x >> 4;
+      const int u = threadIdx.x & 15;
+      const u64 h1 = s_H1[t][u][0] | s_H1[t][u][1] | s_H1[t][u][2] | s_H1[t][u][3];
+      const u64 h0 = s_H0[t][u][0] | s_H0[t][u][1] | s_H0[t][u][2] | s_H0[t][u][3];
+      if (h1 != 0 && h0 != 0) atomicOr(&s_live[t], 1u << u);
+    }
     __syncthreads();
 
     int it = 0;
@@ -804,7 +828,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK, 4) k_scan7_filter(ScanArgs args) {
       }
 
       bool ok = true;
-      for (int u = 0; u < 16 && ok; u++) {
+      for (u32 lm = s_live[t]; lm != 0 && ok; lm &= lm - 1) {
+        const int u = __ffs(lm) - 1;
         u64 acc1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
         u64 acc0[8] = {0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
