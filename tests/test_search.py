@@ -275,3 +275,14 @@ def test_find_circuit_api(tmp_path):
     assert st2.outputs[0] >= 0
     st3 = find_circuit([1, 0, 2, 3], bit=1, seed=5, gpu="off")
     assert st3.outputs[1] >= 0
+
+
+def test_find_circuit_all_outputs(tmp_path, monkeypatch):
+    """find_circuit with bit=None runs the beam search and returns a fully
+    wired, validated state."""
+    monkeypatch.chdir(tmp_path)
+    from sboxgates_amd.search import find_circuit
+    st = find_circuit("des_s1", lut=False, seed=6, gpu="off",
+                      save_dir=str(tmp_path))
+    wired = [b for b in range(8) if st.outputs[b] >= 0]
+    assert len(wired) == 4
