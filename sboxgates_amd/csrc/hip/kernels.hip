@@ -982,11 +982,11 @@ std::unique_ptr<GpuEngine> GpuEngine::create(int device, std::string* err) {
     SBG_HIP_CHECK(hipHostMalloc(&impl->h_ctl, sizeof(DevCtl)));
     SBG_HIP_CHECK(hipHostMalloc(&impl->h_pool, sizeof(ttable) * MAX_GATES));
     // Hit buffer for the 7-LUT frontier: default 16M hits (768 MB) per
-    // chunk; overridable for memory-constrained runs.
+    // chunk; overridable for memory-constrained runs. Allocated lazily on
+    // the first 7-LUT scan (gate-mode engines never need it).
     const char* cap_env = std::getenv("SBOXGATES_HIT_CAP");
     impl->hit_cap = cap_env != nullptr ? std::strtoull(cap_env, nullptr, 10)
                                        : (1ULL << 24);
-    SBG_HIP_CHECK(hipMalloc(&impl->d_hits, sizeof(Hit7) * impl->hit_cap));
     return std::unique_ptr<GpuEngine>(new GpuEngine(impl.release()));
   } catch (const std::exception& ex) {
     if (err != nullptr) *err = ex.what();
@@ -1077,6 +1077,11 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
                                  hipMemcpyDeviceToHost, im->stream));
     SBG_HIP_CHECK(hipStreamSynchronize(im->stream));
   } else if (k == 7) {
+    if (im->d_hits == nullptr) {
+      SBG_HIP_CHECK(hipMalloc(&im->d_hits, sizeof(Hit7) * im->hit_cap));
+      // args was built before the allocation; refresh the pointer.
+    }
+    args.hits = im->d_hits;
     // Filter + assign, with overflow-driven range splitting.
     i64 lo = begin;
     while (lo < end) {

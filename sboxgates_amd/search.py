@@ -32,15 +32,13 @@ def find_circuit(sbox: Union[str, Sequence[int]], bit: Optional[int] = None,
     else:
         table, n = models.load_table(sbox, permute)
 
-    eng = make_engine(lut_graph=lut, seed=seed, gpu=gpu,
-                      oneoutput=-1 if bit is None else bit,
-                      iterations=iterations, metric=metric, try_nots=try_nots,
-                      gate_bitfield=gate_bitfield,
-                      save_states=save_dir is not None,
-                      output_dir=save_dir or "", verbosity=verbosity)
-    eng.set_sbox(table, n)
-
     if bit is not None:
+        eng = make_engine(lut_graph=lut, seed=seed, gpu=gpu, oneoutput=bit,
+                          iterations=iterations, metric=metric,
+                          try_nots=try_nots, gate_bitfield=gate_bitfield,
+                          save_states=save_dir is not None,
+                          output_dir=save_dir or "", verbosity=verbosity)
+        eng.set_sbox(table, n)
         st = eng.initial_state()
         best = None
         for _ in range(iterations):
