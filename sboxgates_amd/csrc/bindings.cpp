@@ -367,6 +367,7 @@ PYBIND11_MODULE(_core, m) {
         return o;
       }))
       .def_readwrite("iterations", &options::iterations)
+      .def_readwrite("jobs", &options::jobs)
       .def_readwrite("oneoutput", &options::oneoutput)
       .def_readwrite("permute", &options::permute)
       .def_readwrite("lut_graph", &options::lut_graph)

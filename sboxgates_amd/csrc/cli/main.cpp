@@ -69,7 +69,10 @@ void print_help(const char* prog) {
       "      --gpus=N                  Drive N GPUs from this process (one\n"
       "                                worker thread per device).\n"
       "      --beam=N                  Tied-state beam width for multi-output\n"
-      "                                search (default 20).\n\n"
+      "                                search (default 20).\n"
+      "      --jobs=N                  Run -i iterations as N parallel jobs\n"
+      "                                (single-output mode; jobs rotate over\n"
+      "                                visible GPUs).\n\n"
       "  -?, --help                    Give this help list.\n"
       "  -V, --version                 Print program version.\n",
       prog);
@@ -91,7 +94,7 @@ int main(int argc, char** argv) {
   opt.set_avail_gates(sbg::DEFAULT_GATE_BITFIELD);
 
   enum { OPT_HIP = 1000, OPT_SEED, OPT_CPU, OPT_GPU, OPT_OUTDIR, OPT_HELP,
-         OPT_GPUS, OPT_BEAM };
+         OPT_GPUS, OPT_BEAM, OPT_JOBS };
   static const struct option long_opts[] = {
       {"available-gates", required_argument, nullptr, 'a'},
       {"convert-c", no_argument, nullptr, 'c'},
@@ -111,6 +114,7 @@ int main(int argc, char** argv) {
       {"output-dir", required_argument, nullptr, OPT_OUTDIR},
       {"gpus", required_argument, nullptr, OPT_GPUS},
       {"beam", required_argument, nullptr, OPT_BEAM},
+      {"jobs", required_argument, nullptr, OPT_JOBS},
       {"help", no_argument, nullptr, OPT_HELP},
       {"version", no_argument, nullptr, 'V'},
       {nullptr, 0, nullptr, 0}};
@@ -169,6 +173,11 @@ int main(int argc, char** argv) {
         v = std::strtol(optarg, &endptr, 10);
         if (*endptr != '\0' || v < 1 || v > 20) return fail("Bad --beam value", optarg);
         opt.beam = static_cast<int>(v);
+        break;
+      case OPT_JOBS:
+        v = std::strtol(optarg, &endptr, 10);
+        if (*endptr != '\0' || v < 1 || v > 64) return fail("Bad --jobs value", optarg);
+        opt.jobs = static_cast<int>(v);
         break;
       case 'V': std::printf("%s\n", kVersion); return 0;
       case OPT_HELP: print_help(argv[0]); return 0;

@@ -5,6 +5,8 @@
 
 #include <algorithm>
 #include <cassert>
+#include <thread>
+#include <vector>
 #include <chrono>
 #include <climits>
 #include <cstdio>
@@ -766,6 +768,10 @@ void Engine::save_checkpoint(const state& st) {
 }
 
 void Engine::generate_graph_one_output(const state& st_in) {
+  if (opt_.jobs > 1 && ctx_->world() == 1) {
+    generate_graph_one_output_jobs(st_in);
+    return;
+  }
   // Parity: sboxgates.c:661-688.
   assert(opt_.iterations > 0);
   state st = st_in;
@@ -794,6 +800,76 @@ void Engine::generate_graph_one_output(const state& st_in) {
     } else {
       if (nst.sat_metric < st.max_sat_metric) st.max_sat_metric = nst.sat_metric;
     }
+  }
+}
+
+// Parallel independent iterations: batches of `jobs` threads, each with
+// its own engine (rotating over visible GPUs when present); search bounds
+// tighten between batches. Semantics: the same iteration count as the
+// serial driver, with bound tightening at batch granularity instead of
+// per-iteration — every produced circuit is identical in kind and
+// checkpointed the same way.
+void Engine::generate_graph_one_output_jobs(const state& st_in) {
+  assert(opt_.iterations > 0);
+  state st = st_in;
+  if (opt_.verbosity >= 0) {
+    std::printf("Generating graphs for output %d (%d parallel jobs)...\n",
+                opt_.oneoutput, opt_.jobs);
+  }
+  const int devices = gpu_ != nullptr ? std::max(1, gpu_count()) : 0;
+  int done = 0;
+  while (done < opt_.iterations) {
+    const int batch = std::min(opt_.jobs, opt_.iterations - done);
+    std::vector<state> results(batch);
+    std::vector<char> found(batch, 0);
+    std::vector<std::thread> threads;
+    for (int j = 0; j < batch; j++) {
+      threads.emplace_back([&, j] {
+        options wopt = opt_;
+        wopt.jobs = 1;
+        wopt.verbosity = -1;
+        if (wopt.seeded) wopt.seed = opt_.seed + 0x9E37 * (done + j + 1);
+        if (devices > 0) wopt.gpu_device = j % devices;
+        try {
+          Engine we(wopt);
+          we.set_sbox(sbox_, num_inputs_);
+          state nst = st;
+          i8 bits[8] = {-1, -1, -1, -1, -1, -1, -1, -1};
+          const ttable mask = tt_mask_for_inputs(get_num_inputs(&st));
+          gatenum out = we.create_circuit(&nst, g_target_[opt_.oneoutput], mask,
+                                          bits);
+          if (out != NO_GATE) {
+            nst.outputs[opt_.oneoutput] = out;
+            results[j] = nst;
+            found[j] = 1;
+          }
+        } catch (const std::exception& e) {
+          std::fprintf(stderr, "search job %d failed: %s\n", done + j, e.what());
+        }
+      });
+    }
+    for (auto& t : threads) t.join();
+    for (int j = 0; j < batch; j++) {
+      if (!found[j]) {
+        if (opt_.verbosity >= 0) {
+          std::printf("(%d/%d): Not found.\n", done + j + 1, opt_.iterations);
+        }
+        continue;
+      }
+      const state& nst = results[j];
+      if (opt_.verbosity >= 0) {
+        std::printf("(%d/%d): %d gates. SAT metric: %d\n", done + j + 1,
+                    opt_.iterations, nst.num_gates - get_num_inputs(&nst),
+                    nst.sat_metric);
+      }
+      save_checkpoint(nst);
+      if (opt_.metric == METRIC_GATES) {
+        if (nst.num_gates < st.max_gates) st.max_gates = nst.num_gates;
+      } else {
+        if (nst.sat_metric < st.max_sat_metric) st.max_sat_metric = nst.sat_metric;
+      }
+    }
+    done += batch;
   }
 }
 

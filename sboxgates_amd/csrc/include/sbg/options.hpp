@@ -44,6 +44,9 @@ struct options {
   int num_gpus = 1;         // CLI --gpus: in-process devices (threads)
   int beam = 20;            // --beam: tied-state beam width (<= 20; the
                             // reference hard-codes 20, sboxgates.c:704)
+  int jobs = 1;             // --jobs: parallel independent search
+                            // iterations (one engine per job; jobs rotate
+                            // over visible GPUs)
   std::string output_dir;   // where XML checkpoints are written ("" = CWD)
   bool save_states = true;  // library callers may disable checkpoint writes
 

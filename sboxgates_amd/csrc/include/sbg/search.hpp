@@ -54,6 +54,9 @@ class Engine {
   void initial_state(state& st) const { init_state(st, num_inputs_); }
 
   // Search drivers (rank 0). Parity: sboxgates.c:661-688, 701-788.
+  // With opt.jobs > 1 (and no distributed ctx), one-output iterations run
+  // as parallel independent jobs, each with its own engine/GPU; bounds
+  // tighten between batches of `jobs` iterations.
   void generate_graph_one_output(const state& st);
   void generate_graph(const state& st);
 
@@ -86,6 +89,7 @@ class Engine {
   // Symmetric distributed 5/7 search body executed by every rank.
   bool distributed_lut_body(const WorkMsg& work, u16 res[10], bool* found5);
   bool dist_scan_chunked(int k, const ScanRequest& rq, u64 chunk, u16 res[10]);
+  void generate_graph_one_output_jobs(const state& st);
   void save_checkpoint(const state& st);
 
   options opt_;

@@ -35,7 +35,7 @@ from .._core import (  # noqa: F401
 def make_engine(lut_graph=False, seed=None, gpu="auto", oneoutput=-1,
                 iterations=1, metric="gates", try_nots=False,
                 save_states=False, output_dir="", verbosity=-1,
-                gate_bitfield=None, ctx=None):
+                gate_bitfield=None, ctx=None, jobs=1):
     """Builds an Engine with the given search options."""
     o = _core.Options()
     o.lut_graph = lut_graph
@@ -45,6 +45,7 @@ def make_engine(lut_graph=False, seed=None, gpu="auto", oneoutput=-1,
     o.gpu = gpu
     o.oneoutput = oneoutput
     o.iterations = iterations
+    o.jobs = jobs
     o.metric = metric
     o.try_nots = try_nots
     o.save_states = save_states

@@ -154,3 +154,11 @@ def test_beam_option(tmp_path):
     assert r.returncode == 0, r.stderr
     assert run(["--beam", "0", des()]).returncode != 0
     assert run(["--beam", "21", des()]).returncode != 0
+
+
+def test_jobs_option(tmp_path):
+    r = run(["-l", "-o", "0", "-i", "4", "--jobs", "2", "--seed", "3", "--cpu",
+             des()], cwd=str(tmp_path), timeout=300)
+    assert r.returncode == 0, r.stderr
+    assert glob.glob(os.path.join(str(tmp_path), "1-*.xml"))
+    assert run(["--jobs", "0", des()]).returncode != 0

@@ -286,3 +286,19 @@ def test_find_circuit_all_outputs(tmp_path, monkeypatch):
                       save_dir=str(tmp_path))
     wired = [b for b in range(8) if st.outputs[b] >= 0]
     assert len(wired) == 4
+
+
+def test_parallel_jobs_one_output(tmp_path):
+    """--jobs: iterations as parallel independent engines; every produced
+    checkpoint is a valid circuit."""
+    sbox, n = models.load("des_s1")
+    eng = make_engine(lut_graph=True, seed=33, gpu="off", save_states=True,
+                      output_dir=str(tmp_path), oneoutput=0, iterations=6,
+                      jobs=3)
+    eng.set_sbox(sbox, n)
+    eng.generate_graph_one_output(eng.initial_state())
+    files = eng.saved_files()
+    assert len(files) >= 2
+    for f in files:
+        st = _core.State.load(f)
+        assert validate_circuit(st, sbox, n, bit=0)
