@@ -70,6 +70,8 @@ void print_help(const char* prog) {
       "                                worker thread per device).\n"
       "      --beam=N                  Tied-state beam width for multi-output\n"
       "                                search (default 20).\n"
+      "      --max-gates=N             Bound the search to N total gates\n"
+      "                                (including inputs).\n"
       "      --jobs=N                  Run -i iterations as N parallel jobs\n"
       "                                (single-output mode; jobs rotate over\n"
       "                                visible GPUs).\n\n"
@@ -94,7 +96,7 @@ int main(int argc, char** argv) {
   opt.set_avail_gates(sbg::DEFAULT_GATE_BITFIELD);
 
   enum { OPT_HIP = 1000, OPT_SEED, OPT_CPU, OPT_GPU, OPT_OUTDIR, OPT_HELP,
-         OPT_GPUS, OPT_BEAM, OPT_JOBS };
+         OPT_GPUS, OPT_BEAM, OPT_JOBS, OPT_MAXG };
   static const struct option long_opts[] = {
       {"available-gates", required_argument, nullptr, 'a'},
       {"convert-c", no_argument, nullptr, 'c'},
@@ -115,6 +117,7 @@ int main(int argc, char** argv) {
       {"gpus", required_argument, nullptr, OPT_GPUS},
       {"beam", required_argument, nullptr, OPT_BEAM},
       {"jobs", required_argument, nullptr, OPT_JOBS},
+      {"max-gates", required_argument, nullptr, OPT_MAXG},
       {"help", no_argument, nullptr, OPT_HELP},
       {"version", no_argument, nullptr, 'V'},
       {nullptr, 0, nullptr, 0}};
@@ -122,6 +125,7 @@ int main(int argc, char** argv) {
   int ch;
   char* endptr = nullptr;
   long v;
+  int max_gates_bound = -1;  // --max-gates: bound on TOTAL gates (incl. inputs)
   while ((ch = getopt_long(argc, argv, "a:cdg:i:lno:p:svV", long_opts, nullptr)) != -1) {
     switch (ch) {
       case 'a':
@@ -178,6 +182,13 @@ int main(int argc, char** argv) {
         v = std::strtol(optarg, &endptr, 10);
         if (*endptr != '\0' || v < 1 || v > 64) return fail("Bad --jobs value", optarg);
         opt.jobs = static_cast<int>(v);
+        break;
+      case OPT_MAXG:
+        v = std::strtol(optarg, &endptr, 10);
+        if (*endptr != '\0' || v < 1 || v > 500) {
+          return fail("Bad --max-gates value", optarg);
+        }
+        max_gates_bound = static_cast<int>(v);
         break;
       case 'V': std::printf("%s\n", kVersion); return 0;
       case OPT_HELP: print_help(argv[0]); return 0;
@@ -286,6 +297,9 @@ int main(int argc, char** argv) {
       return fail("Error when reading state file", err.c_str());
     } else {
       std::printf("Loaded %s.\n", opt.gfname.c_str());
+    }
+    if (max_gates_bound > 0) {
+      st.max_gates = static_cast<sbg::gatenum>(max_gates_bound);
     }
 
     if (opt.oneoutput != -1) {

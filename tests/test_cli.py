@@ -162,3 +162,13 @@ def test_jobs_option(tmp_path):
     assert r.returncode == 0, r.stderr
     assert glob.glob(os.path.join(str(tmp_path), "1-*.xml"))
     assert run(["--jobs", "0", des()]).returncode != 0
+
+
+def test_max_gates_bound(tmp_path):
+    # 6 inputs + 19 gates: usually fails for des bit0 (the bound prunes);
+    # a generous bound succeeds.
+    r = run(["-o", "0", "--seed", "2", "--cpu", "--max-gates", "60", des()],
+            cwd=str(tmp_path))
+    assert r.returncode == 0
+    assert glob.glob(os.path.join(str(tmp_path), "1-*.xml"))
+    assert run(["--max-gates", "0", des()]).returncode != 0
