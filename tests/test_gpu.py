@@ -328,3 +328,22 @@ def test_random_window_count_parity(engines):
             f_c, _, ev_c = cpu.scan_pool(k, st, target, mask, a, b,
                                          count_all=True)
             assert ev_g == ev_c == b - a, (k, a, b, ev_g, ev_c)
+
+
+def test_parallel_jobs_on_gpu():
+    """--jobs: two concurrent engines on one device, each running GPU
+    kernels."""
+    import subprocess
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cli = os.path.join(repo, "bin", "sboxgates")
+    import tempfile
+    with tempfile.TemporaryDirectory() as d:
+        r = subprocess.run(
+            [cli, "-l", "-o", "0", "-i", "4", "--jobs", "2", "--gpu",
+             "--seed", "5", os.path.join(repo, "sboxgates_amd", "sboxes",
+                                         "des_s1.txt")],
+            cwd=d, capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, r.stderr
+        import glob as g
+        assert g.glob(os.path.join(d, "1-*.xml"))
