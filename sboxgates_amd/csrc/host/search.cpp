@@ -882,11 +882,19 @@ static int count_state_outputs(const state& st) {
 }
 
 void Engine::generate_graph(const state& st_in) {
-  // Multi-output beam search, keeping up to 20 tied-minimum start states
-  // per added output (parity: sboxgates.c:701-788).
+  // Multi-output beam search, keeping up to `beam` tied-minimum start
+  // states per added output (parity: sboxgates.c:701-788). With
+  // opt.jobs > 1 (single-process), the (start state x missing output)
+  // searches of one iteration run as parallel independent jobs (engines
+  // rotate over visible GPUs); bounds fold between iterations instead of
+  // between tasks — slightly weaker mid-iteration pruning, same results
+  // semantics.
   int num_start_states = 1;
   state start_states[20];
   start_states[0] = st_in;
+
+  const bool parallel = opt_.jobs > 1 && ctx_->world() == 1;
+  const int devices = gpu_ != nullptr ? std::max(1, gpu_count()) : 0;
 
   int num_outputs;
   while ((num_outputs = count_state_outputs(start_states[0])) < num_outputs_) {
@@ -895,70 +903,135 @@ void Engine::generate_graph(const state& st_in) {
     state out_states[20];
     int num_out_states = 0;
 
+    auto consider = [&](const state& st) {
+      // Fold a successful search result into the beam.
+      if (opt_.metric == METRIC_GATES) {
+        if (max_gates > st.num_gates) {
+          max_gates = st.num_gates;
+          num_out_states = 0;
+        }
+        if (st.num_gates <= max_gates) {
+          if (num_out_states < std::min(20, opt_.beam)) {
+            out_states[num_out_states++] = st;
+          } else if (opt_.verbosity >= 0) {
+            std::printf("Output state buffer full! Throwing away valid state.\n");
+          }
+        }
+      } else {
+        if (max_sat_metric > st.sat_metric) {
+          max_sat_metric = st.sat_metric;
+          num_out_states = 0;
+        }
+        if (st.sat_metric <= max_sat_metric) {
+          if (num_out_states < std::min(20, opt_.beam)) {
+            out_states[num_out_states++] = st;
+          } else if (opt_.verbosity >= 0) {
+            std::printf("Output state buffer full! Throwing away valid state.\n");
+          }
+        }
+      }
+    };
+
     for (int iter = 0; iter < opt_.iterations; iter++) {
       if (opt_.verbosity >= 0) {
         std::printf("Generating circuits with %d output%s. (%d/%d)\n", num_outputs + 1,
                     num_outputs == 0 ? "" : "s", iter + 1, opt_.iterations);
       }
-      for (int current = 0; current < num_start_states; current++) {
-        start_states[current].max_gates = max_gates;
-        start_states[current].max_sat_metric = max_sat_metric;
 
+      // Collect this iteration's tasks.
+      struct Task {
+        int current;
+        u8 output;
+      };
+      std::vector<Task> tasks;
+      for (int current = 0; current < num_start_states; current++) {
         for (u8 output = 0; output < num_outputs_; output++) {
-          if (start_states[current].outputs[output] != NO_GATE) {
-            if (opt_.verbosity >= 0) std::printf("Skipping output %d.\n", output);
-            continue;
-          }
+          if (start_states[current].outputs[output] != NO_GATE) continue;
+          tasks.push_back({current, output});
+        }
+      }
+
+      if (!parallel) {
+        for (const Task& task : tasks) {
           if (opt_.verbosity >= 0) {
-            std::printf("Generating circuit for output %d...\n", output);
+            std::printf("Generating circuit for output %d...\n", task.output);
           }
           i8 bits[8] = {-1, -1, -1, -1, -1, -1, -1, -1};
-          state st = start_states[current];
+          state st = start_states[task.current];
           if (opt_.metric == METRIC_GATES) {
             st.max_gates = max_gates;
           } else {
             st.max_sat_metric = max_sat_metric;
           }
           const ttable mask = tt_mask_for_inputs(get_num_inputs(&st));
-          st.outputs[output] = create_circuit(&st, g_target_[output], mask, bits);
-          if (st.outputs[output] == NO_GATE) {
+          st.outputs[task.output] = create_circuit(&st, g_target_[task.output], mask, bits);
+          if (st.outputs[task.output] == NO_GATE) {
             if (opt_.verbosity >= 0) {
-              std::printf("No solution for output %d.\n", output);
+              std::printf("No solution for output %d.\n", task.output);
             }
             continue;
           }
-          assert(tt_eq_mask(g_target_[output], st.gates[st.outputs[output]].table, mask));
+          assert(tt_eq_mask(g_target_[task.output],
+                            st.gates[st.outputs[task.output]].table, mask));
           save_checkpoint(st);
-
-          int metric_val = opt_.metric == METRIC_GATES ? st.num_gates : st.sat_metric;
-          int& best_metric_i = max_sat_metric;
-          gatenum& best_metric_g = max_gates;
-          if (opt_.metric == METRIC_GATES) {
-            if (best_metric_g > st.num_gates) {
-              best_metric_g = st.num_gates;
-              num_out_states = 0;
-            }
-            if (st.num_gates <= best_metric_g) {
-              if (num_out_states < std::min(20, opt_.beam)) {
-                out_states[num_out_states++] = st;
-              } else if (opt_.verbosity >= 0) {
-                std::printf("Output state buffer full! Throwing away valid state.\n");
+          consider(st);
+        }
+      } else {
+        // Bounds for every task in this round are fixed at round start.
+        const gatenum round_max_gates = max_gates;
+        const int round_max_sat = max_sat_metric;
+        std::vector<state> results(tasks.size());
+        std::vector<char> ok(tasks.size(), 0);
+        size_t next = 0;
+        while (next < tasks.size()) {
+          const size_t batch = std::min<size_t>(opt_.jobs, tasks.size() - next);
+          std::vector<std::thread> threads;
+          for (size_t j = 0; j < batch; j++) {
+            const size_t ti = next + j;
+            threads.emplace_back([&, ti, j] {
+              const Task& task = tasks[ti];
+              options wopt = opt_;
+              wopt.jobs = 1;
+              wopt.verbosity = -1;
+              if (wopt.seeded) {
+                wopt.seed = opt_.seed + 0x51ED * (iter * 1024 + static_cast<int>(ti) + 1);
               }
-            }
-          } else {
-            if (best_metric_i > st.sat_metric) {
-              best_metric_i = st.sat_metric;
-              num_out_states = 0;
-            }
-            if (st.sat_metric <= best_metric_i) {
-              if (num_out_states < std::min(20, opt_.beam)) {
-                out_states[num_out_states++] = st;
-              } else if (opt_.verbosity >= 0) {
-                std::printf("Output state buffer full! Throwing away valid state.\n");
+              if (devices > 0) wopt.gpu_device = static_cast<int>(j) % devices;
+              try {
+                Engine we(wopt);
+                we.set_sbox(sbox_, num_inputs_);
+                state st = start_states[task.current];
+                if (opt_.metric == METRIC_GATES) {
+                  st.max_gates = round_max_gates;
+                } else {
+                  st.max_sat_metric = round_max_sat;
+                }
+                i8 bits[8] = {-1, -1, -1, -1, -1, -1, -1, -1};
+                const ttable mask = tt_mask_for_inputs(get_num_inputs(&st));
+                st.outputs[task.output] =
+                    we.create_circuit(&st, g_target_[task.output], mask, bits);
+                if (st.outputs[task.output] != NO_GATE) {
+                  results[ti] = st;
+                  ok[ti] = 1;
+                }
+              } catch (const std::exception& e) {
+                std::fprintf(stderr, "beam job (state %d, output %d) failed: %s\n",
+                             task.current, task.output, e.what());
               }
-            }
+            });
           }
-          (void)metric_val;
+          for (auto& t : threads) t.join();
+          next += batch;
+        }
+        for (size_t ti = 0; ti < tasks.size(); ti++) {
+          if (!ok[ti]) {
+            if (opt_.verbosity >= 0) {
+              std::printf("No solution for output %d.\n", tasks[ti].output);
+            }
+            continue;
+          }
+          save_checkpoint(results[ti]);
+          consider(results[ti]);
         }
       }
     }

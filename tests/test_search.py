@@ -302,3 +302,19 @@ def test_parallel_jobs_one_output(tmp_path):
     for f in files:
         st = _core.State.load(f)
         assert validate_circuit(st, sbox, n, bit=0)
+
+
+def test_parallel_jobs_beam_search(tmp_path):
+    """generate_graph with jobs: full multi-output beam with parallel
+    (state x output) tasks; final checkpoint covers all outputs."""
+    sbox, n = models.load("des_s1")
+    eng = make_engine(seed=44, gpu="off", save_states=True,
+                      output_dir=str(tmp_path), jobs=3)
+    eng.set_sbox(sbox, n)
+    eng.generate_graph(eng.initial_state())
+    files = eng.saved_files()
+    assert files
+    final = _core.State.load(files[-1])
+    wired = [b for b in range(8) if final.outputs[b] >= 0]
+    assert len(wired) == 4
+    assert validate_circuit(final, sbox, n)
