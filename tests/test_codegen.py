@@ -133,3 +133,9 @@ def test_no_outputs_rejected():
     st.add_gate(6, 0, 1)
     with pytest.raises(RuntimeError):
         _core.graph_to_source(st, "auto")
+
+
+def test_c_request_on_lut_graph_promotes_to_cuda():
+    st, sbox, n = build_circuit("crypto1_fa", 0, lut=True)
+    src = _core.graph_to_source(st, "c")
+    assert "lop3.b32" in src  # the reference's rule: LUT graphs emit CUDA
