@@ -92,6 +92,7 @@ int fail(const char* msg, const char* arg) {
 }  // namespace
 
 int main(int argc, char** argv) {
+  std::setvbuf(stdout, nullptr, _IOLBF, 0);  // keep logs on SIGKILL'd runs
   sbg::options opt;
   opt.set_avail_gates(sbg::DEFAULT_GATE_BITFIELD);
 
