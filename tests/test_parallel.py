@@ -64,3 +64,11 @@ def test_run_search_driver(tmp_path):
     """The parallel.run_search SPMD helper completes on both ranks."""
     res = launch("run_search_driver", tmp_path)
     assert res[0]["ok"] and res[1]["ok"]
+
+
+def test_distributed_world3_small_chunks(tmp_path):
+    """Three ranks, uneven ranges, forced multi-chunk cadence."""
+    res = launch("one_output_search", tmp_path, world=3,
+                 extra_env={"SBOXGATES_CHUNK5": "1500",
+                            "SBOXGATES_CHUNK7": "15000"})
+    assert all(res[r]["ok"] for r in range(3))
