@@ -318,3 +318,26 @@ def test_parallel_jobs_beam_search(tmp_path):
     wired = [b for b in range(8) if final.outputs[b] >= 0]
     assert len(wired) == 4
     assert validate_circuit(final, sbox, n)
+
+
+def test_gpu_force_without_gpu_raises():
+    if _core.gpu_available():
+        pytest.skip("GPU present")
+    with pytest.raises(RuntimeError):
+        make_engine(gpu="force")
+
+
+@pytest.mark.parametrize("bitfield", [0x8001, 8, 0x6, 1])
+def test_degenerate_gate_sets_no_crash(bitfield):
+    """TRUE/FALSE/pass-through-only vocabularies must terminate cleanly
+    (found or not) rather than loop or crash."""
+    sbox, n = models.load("crypto1_fa")
+    eng = make_engine(seed=2, gpu="off", save_states=False,
+                      gate_bitfield=bitfield)
+    eng.set_sbox(sbox, n)
+    st = eng.initial_state()
+    st.max_gates = 14
+    out = eng.create_circuit(st, eng.target(0), mask_for_inputs(n))
+    if out >= 0:
+        st.set_output(0, out)
+        assert validate_circuit(st, sbox, n, bit=0)
