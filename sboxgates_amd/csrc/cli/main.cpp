@@ -13,6 +13,8 @@
 // multi-process deployment.
 
 #include <getopt.h>
+#include <sys/stat.h>
+#include <sys/types.h>
 
 #include <atomic>
 #include <climits>
@@ -168,7 +170,10 @@ int main(int argc, char** argv) {
         break;
       case OPT_CPU: opt.gpu = sbg::GPU_OFF; break;
       case OPT_GPU: opt.gpu = sbg::GPU_FORCE; break;
-      case OPT_OUTDIR: opt.output_dir = optarg; break;
+      case OPT_OUTDIR:
+        opt.output_dir = optarg;
+        (void)mkdir(optarg, 0755);  // best-effort; open errors surface later
+        break;
       case OPT_GPUS:
         v = std::strtol(optarg, &endptr, 10);
         if (*endptr != '\0' || v < 1 || v > 64) return fail("Bad --gpus value", optarg);
