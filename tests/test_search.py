@@ -341,3 +341,22 @@ def test_degenerate_gate_sets_no_crash(bitfield):
     if out >= 0:
         st.set_output(0, out)
         assert validate_circuit(st, sbox, n, bit=0)
+
+
+def test_scan_windows_at_max_pool():
+    """Boundary math at the MAX_GATES pool size (n=500): window counts are
+    exact at the far end of the combination space."""
+    sbox, n = models.load("rijndael")
+    eng = make_engine(lut_graph=True, seed=1, gpu="off", save_states=False)
+    eng.set_sbox(sbox, n)
+    st = eng.initial_state()
+    st.grow_pool_random(500, 99)
+    assert st.num_gates == 500
+    mask = mask_for_inputs(8)
+    for k in (3, 5, 7):
+        total = n_choose_k(500, k)
+        for begin in (0, total // 2, total - 100_000):
+            end = min(total, begin + 100_000)
+            _, _, ev = eng.scan_pool(k, st, eng.target(0), mask, begin, end,
+                                     count_all=True)
+            assert ev == end - begin, (k, begin)
