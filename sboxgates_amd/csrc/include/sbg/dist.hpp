@@ -51,4 +51,9 @@ struct WorkMsg {
 constexpr i32 WORK_QUIT = 0;
 constexpr i32 WORK_LUT_SEARCH = 1;
 
+// The WorkMsg is broadcast as raw bytes between ranks (same-arch SPMD);
+// lock its ABI so a header change cannot silently desynchronize ranks
+// built from different trees.
+static_assert(sizeof(WorkMsg) == 32160, "WorkMsg wire format changed");
+
 }  // namespace sbg
