@@ -89,6 +89,11 @@ static gatenum add_or_gate(state* st, gatenum g1, gatenum g2, metric_t metric) {
 
 static gatenum add_xor_gate(state* st, gatenum g1, gatenum g2, metric_t metric) {
   if (g1 == NO_GATE || g2 == NO_GATE) return NO_GATE;
+  // XOR(x, x) is constant FALSE — a degenerate multiplexer (both
+  // half-space solutions equal the selector bit). The reference ABORTS
+  // here (add_gate's gid1 != gid2 assert, sboxgates.c:103, reachable for
+  // degenerate targets); skipping the variant lets the search continue.
+  if (g1 == g2) return NO_GATE;
   return add_gate(st, XOR, g1, g2, metric);
 }
 

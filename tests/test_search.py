@@ -360,3 +360,18 @@ def test_scan_windows_at_max_pool():
             _, _, ev = eng.scan_pool(k, st, eng.target(0), mask, begin, end,
                                      count_all=True)
             assert ev == end - begin, (k, begin)
+
+
+def test_degenerate_mux_no_abort():
+    """A target whose both mux half-solutions are the selector bit made
+    the reference abort (XOR(x,x) assert); this engine must complete.
+    (Config found by randomized soak testing.)"""
+    sbox, n = models.load_table([4, 6, 4, 0, 0, 2, 4, 2])
+    eng = make_engine(seed=2108833752, gpu="off", save_states=False,
+                      metric="sat", gate_bitfield=214)
+    eng.set_sbox(sbox, n)
+    st = eng.initial_state()
+    out = eng.create_circuit(st, eng.target(0), mask_for_inputs(n))
+    if out >= 0:
+        st.set_output(0, out)
+        assert validate_circuit(st, sbox, n, bit=0)
