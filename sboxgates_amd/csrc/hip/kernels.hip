@@ -230,8 +230,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan3(ScanArgs args) {
     }
     u32 p1, p0;
     if (lut3_p_masks(ta, tb, tc, args.T1, args.T0, &p1, &p0)) {
-      if (!args.count_all) {
-        u8 func = lut3_function_from_p(p1, p0, dev_rnd(args.seed, idx));
+      u8 func = lut3_function_from_p(p1, p0, dev_rnd(args.seed, idx));
+      if (!args.count_all && func != 0) {
         u16 res[10] = {};
         res[0] = func;
         res[1] = static_cast<u16>(a);

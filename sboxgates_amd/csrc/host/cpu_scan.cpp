@@ -39,7 +39,7 @@ ScanResult cpu_scan3(const ScanRequest& rq, i64 begin, i64 end) {
     if (lut3_p_masks(rq.tables[nums[0]], rq.tables[nums[1]], rq.tables[nums[2]],
                      T1, T0, &p1, &p0)) {
       u8 func = lut3_function_from_p(p1, p0, hash_mix64(rq.seed ^ static_cast<u64>(i)));
-      if (!rq.count_all) {
+      if (func != 0 && !rq.count_all) {
         out.found = true;
         out.res[0] = func;
         out.res[1] = nums[0];

@@ -102,9 +102,17 @@ SBG_HD inline bool lut7_p_masks(const ttable* t, const ttable& T1, const ttable&
 }
 
 // 3-LUT function from cell masks: forced-1 bits, don't-cares randomized.
+// Guarantees a NONZERO byte whenever any don't-care exists: function 00
+// does not round-trip through the gates.xsd loader (the reference's own
+// validator rejects func <= 0, state.c:300 — an upstream latent bug its
+// randomizer can also trigger). A forced-all-zero function returns 0 and
+// callers treat the candidate as a miss (a constant-FALSE LUT is never a
+// useful solution).
 SBG_HD inline u8 lut3_function_from_p(u32 p1, u32 p0, u64 rnd) {
   u8 dontcare = static_cast<u8>(~(p1 | p0) & 0xff);
-  return static_cast<u8>((p1 & 0xff) | (dontcare & static_cast<u8>(rnd)));
+  u8 f = static_cast<u8>((p1 & 0xff) | (dontcare & static_cast<u8>(rnd)));
+  if (f == 0 && dontcare != 0) f = static_cast<u8>(dontcare & (-dontcare));
+  return f;
 }
 
 // ---------------------------------------------------------------------------
@@ -197,6 +205,10 @@ SBG_HD inline bool lut5_solve_from_p(u32 p1, u32 p0, u64 rnd, u8* fo_out,
     for (int u = 0; u < 8; u++) adj[u] &= static_cast<u8>(~(1u << u));
     u8 fo;
     if (!two_color_8(adj, rnd, &fo)) continue;
+    // function 00 does not round-trip (see lut3_function_from_p): the
+    // complement of a proper coloring is proper, so an all-zero outer
+    // byte can always be flipped.
+    if (fo == 0) fo = 0xff;
     // Inner function over cells (o, v): inner pattern = o<<2 | v.
     u8 fi = 0, forced = 0;
     for (int v = 0; v < 4; v++) {
@@ -210,6 +222,14 @@ SBG_HD inline bool lut5_solve_from_p(u32 p1, u32 p0, u64 rnd, u8* fo_out,
       if (A0[v] & sel0) { forced |= 1u << v; }
     }
     fi |= static_cast<u8>(~forced) & static_cast<u8>(rnd >> 24);
+    if (fi == 0) {
+      if (forced != 0xff) {
+        u8 dc = static_cast<u8>(~forced);
+        fi = static_cast<u8>(dc & (-dc));
+      } else {
+        continue;  // inner forced identically 0: degenerate, try next split
+      }
+    }
     *fo_out = fo;
     *fi_out = fi;
     *split_out = s;
@@ -290,6 +310,10 @@ SBG_HD inline bool lut7_solve_ordering(const u64 p1[2], const u64 p0[2],
 
   for (int fmi = fm_offset; fmi < fm_offset + fm_count; fmi++) {
     u8 fm = static_cast<u8>((fmi + (rnd >> 32)) & 0xff);
+    // fm = 00 does not round-trip; a constant-0 middle input is always
+    // expressible as fm = ff with the inner function adapting, which the
+    // sweep reaches anyway.
+    if (fm == 0) continue;
     // Layers for the outer coloring: (m, g) in {0,1}^2.
     // L1[layer] bit u = exists middle-cell w with fm-class m and p1 set.
     u8 L1[4] = {0, 0, 0, 0};
@@ -321,6 +345,7 @@ SBG_HD inline bool lut7_solve_ordering(const u64 p1[2], const u64 p0[2],
     if (!two_color_8(adj, rnd ^ (static_cast<u64>(fm) * 0x9E3779B97F4A7C15ULL), &fo)) {
       continue;
     }
+    if (fo == 0) fo = 0xff;  // complement coloring: function 00 never emitted
     // Inner function over (o, m, g): pattern = o<<2 | m<<1 | g.
     u8 fi = 0, forced = 0;
     for (int m = 0; m < 2; m++) {
@@ -336,6 +361,14 @@ SBG_HD inline bool lut7_solve_ordering(const u64 p1[2], const u64 p0[2],
       }
     }
     fi |= static_cast<u8>(~forced) & static_cast<u8>(rnd >> 16);
+    if (fi == 0) {
+      if (forced != 0xff) {
+        u8 dc = static_cast<u8>(~forced);
+        fi = static_cast<u8>(dc & (-dc));
+      } else {
+        continue;  // inner forced identically 0: degenerate, next fm
+      }
+    }
     *fo_out = fo;
     *fm_out = fm;
     *fi_out = fi;

@@ -151,3 +151,24 @@ def test_naive_check_matches_derivation():
         if ok:
             t = gen_lut_ttable(fun, *tables)
             assert tt_eq_mask(target, t, mask)
+
+
+def test_no_zero_function_bytes_emitted():
+    """Function byte 00 does not round-trip through the gates.xsd loader
+    (the reference's validator rejects it); solvers must never emit it.
+    Constructed degenerate targets force the all-zero corner."""
+    rng = random.Random(99)
+    zero = b"\x00" * 32
+    ones = b"\xff" * 32
+    for trial in range(200):
+        tables = [rand_tt(rng) for _ in range(5)]
+        # Degenerate targets: constant 0 / constant 1 / sparse masks.
+        target = rng.choice([zero, ones, rand_tt(rng)])
+        mask = rng.choice([ones, rand_sparse_tt(rng, 4)])
+        found, fo, fi, split = lut5_solve(tables, target, mask, rng.getrandbits(64))
+        if found:
+            assert fo != 0 and fi != 0, trial
+        t7 = [rand_tt(rng) for _ in range(7)]
+        found7, sol = lut7_solve(t7, target, mask, rng.getrandbits(64))
+        if found7:
+            assert sol[0] != 0 and sol[1] != 0 and sol[2] != 0, trial
