@@ -337,6 +337,8 @@ def test_parallel_jobs_on_gpu():
     import os
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     cli = os.path.join(repo, "bin", "sboxgates")
+    if not os.path.exists(cli):
+        pytest.skip("CLI not built")
     import tempfile
     with tempfile.TemporaryDirectory() as d:
         r = subprocess.run(
