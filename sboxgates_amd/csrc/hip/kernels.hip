@@ -1123,6 +1123,9 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
           throw std::runtime_error("7-LUT hit buffer too small for one combo");
         }
       }
+      // Accounting invariant: h_ctl is zeroed at the top of every attempt,
+      // so only the final (non-overflowed) attempt's counter lands here —
+      // combinations re-scanned after an overflow are never double-counted.
       out.evaluated += im->h_ctl->evaluated;
       unsigned long long nhits = im->h_ctl->hit_count;
       if (nhits > im->hit_cap) nhits = im->hit_cap;

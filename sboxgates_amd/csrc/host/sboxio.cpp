@@ -42,8 +42,26 @@ bool load_sbox_file(const std::string& path, int permute, u8 sbox[256],
   int len = 0;
   unsigned value;
   int ret;
-  while ((ret = std::fscanf(fp, " %x", &value)) > 0 && len < 256 && value < 256) {
+  // Strict parse: an out-of-range entry, a non-hex token, or content beyond
+  // 256 entries is a load error. (A silent stop here could truncate a corrupt
+  // file to a smaller power-of-two length that then loads "successfully".)
+  while ((ret = std::fscanf(fp, " %x", &value)) > 0) {
+    if (value >= 256) {
+      std::fclose(fp);
+      if (err != nullptr) *err = "S-box entry out of range (>= 0x100): " + path;
+      return false;
+    }
+    if (len >= 256) {
+      std::fclose(fp);
+      if (err != nullptr) *err = "more than 256 entries in S-box file: " + path;
+      return false;
+    }
     table[len++] = static_cast<u8>(value);
+  }
+  if (ret == 0) {  // non-hex garbage before EOF
+    std::fclose(fp);
+    if (err != nullptr) *err = "unparseable token in S-box file: " + path;
+    return false;
   }
   std::fclose(fp);
   return load_sbox_table(table, len, permute, sbox, num_inputs, err);

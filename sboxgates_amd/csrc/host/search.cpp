@@ -29,6 +29,11 @@ gatenum add_gate(state* st, int type, gatenum gid1, gatenum gid2, metric_t metri
   assert(!(type == NOT && gid2 != NO_GATE));
   assert(type != IN && type != LUT);
   if (gid1 == NO_GATE || (gid2 == NO_GATE && type != NOT)) return NO_GATE;
+  // Hard capacity check first: the reference's `num_gates > max_gates` test
+  // alone writes one past gates[MAX_GATES-1] when max_gates == MAX_GATES
+  // (the init_state default) and a mux chain reaches exactly MAX_GATES
+  // gates — an out-of-bounds write in the reference; rejected here.
+  if (st->num_gates >= MAX_GATES) return NO_GATE;
   if (st->num_gates > st->max_gates) return NO_GATE;
   if (metric == METRIC_SAT && st->sat_metric > st->max_sat_metric) return NO_GATE;
   assert(gid1 < st->num_gates);
@@ -59,7 +64,7 @@ gatenum add_not_gate(state* st, gatenum gid, metric_t metric) {
 gatenum add_lut(state* st, u8 func, const ttable& table, gatenum g1, gatenum g2,
                 gatenum g3) {
   if (g1 == NO_GATE || g2 == NO_GATE || g3 == NO_GATE ||
-      st->num_gates > st->max_gates) {
+      st->num_gates >= MAX_GATES || st->num_gates > st->max_gates) {
     return NO_GATE;
   }
   assert(g1 < st->num_gates && g2 < st->num_gates && g3 < st->num_gates);
@@ -722,7 +727,7 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
         gatenum fe = create_circuit(&nst_or, nst_or.gates[fd].table ^ target,
                                     mask & ~fsel, next_inbits);
         nst_or.max_gates += 2;
-        nst_or.max_sat_metric += sat_metric_of(AND) + sat_metric_of(XOR);
+        nst_or.max_sat_metric += sat_metric_of(OR) + sat_metric_of(XOR);
         gatenum org = add_or_gate(&nst_or, fe, static_cast<gatenum>(bit), opt_.metric);
         mux_out_or = add_xor_gate(&nst_or, fd, org, opt_.metric);
         nst_or.max_gates = st->max_gates;

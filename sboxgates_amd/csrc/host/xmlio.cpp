@@ -309,9 +309,13 @@ bool state_from_xml(const std::string& xml, state* out, std::string* err) {
     if (funcstr != nullptr) {
       char* end = nullptr;
       func = std::strtol(funcstr, &end, 16);
-      if (end == funcstr || func <= 0 || func > 255) return fail("bad LUT function");
+      // Full-token consumption: trailing garbage ("1g") is an error, not 1.
+      if (end == funcstr || *end != '\0' || func <= 0 || func > 255) {
+        return fail("bad LUT function");
+      }
     }
     if (type != LUT && func != 0) return fail("function on non-LUT gate");
+    if (type == LUT && funcstr == nullptr) return fail("LUT gate without function");
 
     int inp = 0;
     gatenum inputs[3] = {NO_GATE, NO_GATE, NO_GATE};
