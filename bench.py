@@ -38,6 +38,9 @@ def main():
     ap.add_argument("--allow-cpu", action="store_true",
                     help="permit the CPU scan path (dev only; GPU runs must "
                          "use the gfx950 kernels)")
+    ap.add_argument("--no-gate-search", action="store_true",
+                    help="skip the post-timing AES bit-0 LUT search that "
+                         "produces the final-gate-count half of the metric")
     args = ap.parse_args()
 
     import torch
@@ -127,6 +130,22 @@ def main():
         elapsed_max = elapsed
         evaluated_sum = evaluated
 
+    # The BASELINE metric is "LUT-candidates/sec + final gate count": rank 0
+    # additionally runs the actual flagship search (AES S-box bit 0, 3-LUT
+    # graph) end-to-end, outside the timed scan region, and reports the gate
+    # count of the circuit it produced this run (validated by DAG evaluation
+    # inside find_circuit). Reference quality bar: the reference's
+    # illustrative single-output AES LUT artifact is 67 gates
+    # (/root/reference/README.md:107-112).
+    final_gate_count = None
+    gate_search_seconds = None
+    if rank == 0 and not args.no_gate_search and use_gpu:
+        from sboxgates_amd.search import find_circuit
+        ts = time.perf_counter()
+        best = find_circuit("rijndael", bit=0, lut=True, seed=11, gpu="force")
+        gate_search_seconds = time.perf_counter() - ts
+        final_gate_count = best.num_gates - 8
+
     if rank == 0:
         value = evaluated_sum / elapsed_max
         out = {
@@ -142,6 +161,8 @@ def main():
             "vs_baseline": None,
             "dtype": "uint64-bitslice",
             "data": "synthetic",
+            "final_gate_count": final_gate_count,
+            "gate_search_seconds": gate_search_seconds,
             "config": {
                 "model": "rijndael S-box, output bit 0, 5-LUT exhaustive scan",
                 "pool_gates": args.pool_gates,

@@ -94,33 +94,47 @@ def test_planted_solutions_found_and_valid(engines, k):
 
 def test_found_parity_sparse_masks(engines):
     """GPU and CPU must agree on found/not-found for sparse-mask scans
-    (where real hits occur) across many random instances."""
+    (deep-recursion-like conditions) across many random instances.
+
+    Half the instances plant a 5-LUT-decomposable target (realizable under
+    ANY mask), so a hit is guaranteed and a regression that kills hit
+    detection cannot hide behind not-found agreement; the other half use
+    raw S-box output-bit targets for unbiased found/not-found parity."""
     gpu, cpu = engines
+    import struct
     rng = random.Random(31)
-    agree_found = 0
+    found_planted = 0
     for trial in range(20):
         pool = rng.choice([20, 30, 40])
         st = make_pool(gpu, pool, rng.getrandbits(32))
         # Sparse mask: search deep-recursion-like conditions.
-        import struct
         words = [0, 0, 0, 0]
         for _ in range(rng.choice([6, 10, 16])):
             i = rng.randrange(256)
             words[i // 64] |= 1 << (i % 64)
         mask = struct.pack("<4Q", *words)
-        target = gpu.target(rng.randrange(8))
+        planted = trial % 2 == 0
+        if planted:
+            ids = rng.sample(range(pool), 5)
+            tabs = [st.gate(i)["table"] for i in ids]
+            t_o = gen_lut_ttable(rng.randrange(256), tabs[0], tabs[1], tabs[2])
+            target = gen_lut_ttable(rng.randrange(256), t_o, tabs[3], tabs[4])
+        else:
+            target = gpu.target(rng.randrange(8))
         total = n_choose_k(pool, 5)
         f_g, r_g, _ = gpu.scan_pool(5, st, target, mask, 0, total, seed=trial)
         f_c, r_c, _ = cpu.scan_pool(5, st, target, mask, 0, total, seed=trial)
         assert f_g == f_c, trial
+        if planted:
+            assert f_g, f"planted 5-LUT not found in trial {trial}"
+            found_planted += 1
         if f_g:
-            agree_found += 1
             t_o = gen_lut_ttable(r_g[0], st.gate(r_g[2])["table"],
                                  st.gate(r_g[3])["table"], st.gate(r_g[4])["table"])
             got = gen_lut_ttable(r_g[1], t_o, st.gate(r_g[5])["table"],
                                  st.gate(r_g[6])["table"])
             assert tt_eq_mask(target, got, mask)
-    assert agree_found >= 3
+    assert found_planted == 10
 
 
 def test_range_split_counts(engines):
@@ -328,6 +342,67 @@ def test_random_window_count_parity(engines):
             f_c, _, ev_c = cpu.scan_pool(k, st, target, mask, a, b,
                                          count_all=True)
             assert ev_g == ev_c == b - a, (k, a, b, ev_g, ev_c)
+
+
+def test_gpu_window_fuzz_soak(engines):
+    """Randomized GPU-vs-CPU fuzz over (k, pool, window, mask sparsity):
+    evaluated counts must match exactly, found verdicts must agree, and any
+    hit must verify under the mask. This is the GPU-path analog of the CPU
+    fuzz soaks (round-1 soaks exercised only the CPU path)."""
+    import struct
+    gpu, cpu = engines
+    rng = random.Random(0xF022)
+    for trial in range(60):
+        k = rng.choice([3, 4, 5, 5, 7, 7])
+        pool = rng.choice([15, 25, 40, 70])
+        if k == 7 and pool > 40:
+            pool = 40
+        st = make_pool(gpu, pool, rng.getrandbits(32))
+        # Mask: dense, sparse, or full.
+        kind = rng.randrange(3)
+        if kind == 0:
+            mask = mask_for_inputs(8)
+        else:
+            words = [0, 0, 0, 0]
+            nbits = rng.choice([4, 12, 40, 150])
+            for _ in range(nbits):
+                i = rng.randrange(256)
+                words[i // 64] |= 1 << (i % 64)
+            mask = struct.pack("<4Q", *words)
+        target = gpu.target(rng.randrange(8))
+        total = n_choose_k(pool, 3 if k == 4 else k)
+        a = rng.randrange(total)
+        b = min(total, a + rng.choice([3, 500, 30_000, 200_000]))
+        f_g, r_g, ev_g = gpu.scan_pool(k, st, target, mask, a, b,
+                                       seed=trial, count_all=True)
+        f_c, r_c, ev_c = cpu.scan_pool(k, st, target, mask, a, b,
+                                       seed=trial, count_all=True)
+        assert ev_g == ev_c == b - a, (trial, k, pool, a, b, ev_g, ev_c)
+        # Early-exit mode: verdicts agree; hits verify.
+        f_g, r_g, _ = gpu.scan_pool(k, st, target, mask, a, b, seed=trial)
+        f_c, _, _ = cpu.scan_pool(k, st, target, mask, a, b, seed=trial)
+        assert f_g == f_c, (trial, k, pool, a, b)
+        if f_g and k in (3, 5, 7):
+            if k == 3:
+                got = gen_lut_ttable(r_g[0], st.gate(r_g[1])["table"],
+                                     st.gate(r_g[2])["table"],
+                                     st.gate(r_g[3])["table"])
+            elif k == 5:
+                t_o = gen_lut_ttable(r_g[0], st.gate(r_g[2])["table"],
+                                     st.gate(r_g[3])["table"],
+                                     st.gate(r_g[4])["table"])
+                got = gen_lut_ttable(r_g[1], t_o, st.gate(r_g[5])["table"],
+                                     st.gate(r_g[6])["table"])
+            else:
+                t_o = gen_lut_ttable(r_g[0], st.gate(r_g[3])["table"],
+                                     st.gate(r_g[4])["table"],
+                                     st.gate(r_g[5])["table"])
+                t_m = gen_lut_ttable(r_g[1], st.gate(r_g[6])["table"],
+                                     st.gate(r_g[7])["table"],
+                                     st.gate(r_g[8])["table"])
+                got = gen_lut_ttable(r_g[2], t_o, t_m,
+                                     st.gate(r_g[9])["table"])
+            assert tt_eq_mask(target, got, mask)
 
 
 def test_parallel_jobs_on_gpu():
