@@ -26,11 +26,17 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <chrono>
+#include <cstddef>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <map>
+#include <memory>
+#include <mutex>
 #include <stdexcept>
 #include <string>
+#include <thread>
 #include <vector>
 
 #include "sbg/comb.hpp"
@@ -272,33 +278,20 @@ __device__ __forceinline__ u8 dev_permute_cells8(u8 m, int s0, int s1, int s2) {
 __constant__ int c_perm6[6][3] = {{0, 1, 2}, {0, 2, 1}, {1, 0, 2},
                                   {1, 2, 0}, {2, 0, 1}, {2, 1, 0}};
 
-__global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
-  __shared__ alignas(16) u64 s_pool[MAX_GATES * PSTR];
-  __shared__ alignas(16) u8 s_bitmap[256 * 256 / 8];
-  __shared__ u8 s_funs[256];
-  __shared__ int s_count;
-  const int n = args.n;
-  load_pool_lds(s_pool, args.pool, n);
-  for (int i = threadIdx.x; i < 256 * 256 / 8 / 8; i += blockDim.x) {
-    reinterpret_cast<u64*>(s_bitmap)[i] =
-        reinterpret_cast<const u64*>(args.matcher->bitmap)[i];
-  }
-  for (int i = threadIdx.x; i < 256 / 8; i += blockDim.x) {
-    reinterpret_cast<u64*>(s_funs)[i] =
-        reinterpret_cast<const u64*>(args.matcher->funs)[i];
-  }
-  if (threadIdx.x == 0) s_count = args.matcher->count;
-  __syncthreads();
-
-  DevCtl* ctl = args.ctl;
+// Shared k=4 scan loop: grid-strided walk of triple ranks [idx0, end) with
+// stride `stride`, pool/matcher staged in LDS by the caller. Returns this
+// thread's evaluated count. Used by both the one-shot k_scan4 kernel and
+// the persistent scan service below.
+__device__ __forceinline__ u64 scan4_loop(const u64* s_pool, const u8* s_bitmap,
+                                          const u8* s_funs, int s_count, int n,
+                                          const ttable& T1, const ttable& T0,
+                                          int count_all, DevCtl* ctl, i64 idx0,
+                                          i64 end, i64 stride) {
   const i64 total3 = cf3(n);
   u64 local_eval = 0;
-  const i64 stride = static_cast<i64>(gridDim.x) * blockDim.x;
-  i64 idx = args.begin + blockIdx.x * static_cast<i64>(blockDim.x) + threadIdx.x;
   int tick = 0;
-
-  for (; idx < args.end; idx += stride) {
-    if (((tick++) & 255) == 0 && !args.count_all && dev_abort(ctl)) break;
+  for (i64 idx = idx0; idx < end; idx += stride) {
+    if (((tick++) & 255) == 0 && !count_all && dev_abort(ctl)) break;
     int a = first_of_rank<cf3>(idx, n, total3);
     i64 rem = idx - (total3 - cf3(n - a));
     int b2, c2;
@@ -314,8 +307,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
       tc.w[w] = s_pool[c * PSTR + w];
     }
     u32 p1, p0;
-    if (!lut3_p_masks(ta, tb, tc, args.T1, args.T0, &p1, &p0)) continue;
-    if (args.count_all) continue;
+    if (!lut3_p_masks(ta, tb, tc, T1, T0, &p1, &p0)) continue;
+    if (count_all) continue;
     const u8 req1 = static_cast<u8>(p1);
     const u8 care = static_cast<u8>(p1 | p0);
     for (int perm = 0; perm < 6; perm++) {
@@ -339,8 +332,36 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
       }
       break;  // bitmap said a function exists; it was found and published
     }
-    if (dev_abort(ctl) && !args.count_all) break;
+    if (dev_abort(ctl) && !count_all) break;
   }
+  return local_eval;
+}
+
+__global__ void __launch_bounds__(SCAN_BLOCK) k_scan4(ScanArgs args) {
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * PSTR];
+  __shared__ alignas(16) u8 s_bitmap[256 * 256 / 8];
+  __shared__ u8 s_funs[256];
+  __shared__ int s_count;
+  const int n = args.n;
+  load_pool_lds(s_pool, args.pool, n);
+  for (int i = threadIdx.x; i < 256 * 256 / 8 / 8; i += blockDim.x) {
+    reinterpret_cast<u64*>(s_bitmap)[i] =
+        reinterpret_cast<const u64*>(args.matcher->bitmap)[i];
+  }
+  for (int i = threadIdx.x; i < 256 / 8; i += blockDim.x) {
+    reinterpret_cast<u64*>(s_funs)[i] =
+        reinterpret_cast<const u64*>(args.matcher->funs)[i];
+  }
+  if (threadIdx.x == 0) s_count = args.matcher->count;
+  __syncthreads();
+
+  DevCtl* ctl = args.ctl;
+  const i64 stride = static_cast<i64>(gridDim.x) * blockDim.x;
+  const i64 idx0 =
+      args.begin + blockIdx.x * static_cast<i64>(blockDim.x) + threadIdx.x;
+  u64 local_eval = scan4_loop(s_pool, s_bitmap, s_funs, s_count, n, args.T1,
+                              args.T0, args.count_all, ctl, idx0, args.end,
+                              stride);
   __shared__ unsigned long long s_eval;
   if (threadIdx.x == 0) s_eval = 0;
   __syncthreads();
@@ -916,7 +937,532 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan7_assign(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Persistent k=4 scan service.
+//
+// Gate-mode searches issue one step-4 triple scan per recursion node
+// (~600k nodes for AES bit 0); through the one-shot launch path each scan
+// pays a ~30 us API floor (pool H2D + ctl round trips + launch + sync) that
+// dwarfs the kernel time of small scans. The service keeps a grid of
+// resident workgroups parked on a mailbox in fine-grained pinned host
+// memory: the host publishes {request header, pool delta} with one release
+// store, workgroup 0 copies the delta into the device pool and re-publishes
+// device-side, and the last workgroup to finish writes the response
+// directly back to pinned memory — no launches, no stream ops, ~5-8 us per
+// scan. (Price anchors: MI355X_MICROARCH.md persistent-kernel price list —
+// host-paired flag round trips are single-digit us; relaxed polls +
+// s_sleep from one lane, one acquire fence on the hit.)
+//
+// Liveness/safety design (every wait is bounded):
+//  * The kernel retires itself after ~2 ms idle (svc_state -> RETIRED) so
+//    an idle service never starves other kernels of CUs; the host reaps
+//    and relaunches on the next submit (and resolves the publish/retire
+//    race by relaunching when it observes RETIRED while waiting).
+//  * Residency self-check: workgroups count themselves in at startup; if
+//    the grid is not fully resident within a deadline the host quits the
+//    kernel and permanently falls back to the one-shot path.
+//  * The host can always stop the kernel without its cooperation: leader
+//    workgroup polls quit_req (host memory); the other workgroups poll
+//    dev_seq (device memory) which the host can overwrite via an async
+//    copy on a separate stream, and the scan loops poll ctl.abort which
+//    the host can set the same way.
+// ---------------------------------------------------------------------------
+
+enum : u32 { SVC_RUNNING = 1u, SVC_RETIRED = 2u };
+enum : u32 { SVC_CMD_WORK = 0u, SVC_CMD_QUIT = 1u };
+
+// Request header word layout (u64[16] block, read cooperatively).
+enum {
+  HDR_N_KEEP = 0,     // n (lo32) | pool_keep (hi32)
+  HDR_EPOCH_CALL = 1, // matcher_epoch (lo32) | count_all (hi32)
+  HDR_BEGIN = 2,
+  HDR_END = 3,
+  HDR_SEED = 4,
+  HDR_T1 = 8,         // 8..11
+  HDR_T0 = 12,        // 12..15
+  HDR_WORDS = 16,
+};
+
+struct SvcMailbox {  // pinned fine-grained host memory
+  // host -> device control
+  u64 req_seq;       // release-stored by the host to publish a request
+  u32 quit_req;
+  u32 svc_state;     // SVC_RUNNING / SVC_RETIRED
+  // device -> host
+  u64 resp_seq;      // release-stored by the last workgroup
+  u64 r_evaluated;
+  u32 r_found;
+  u32 alive;         // residency self-check counter
+  u16 r_res[10];
+  u16 pad[2];
+  // request payload
+  alignas(64) u64 hdr[HDR_WORDS];
+  alignas(64) ttable pool_staging[MAX_GATES];  // full pool image (host shadow)
+  alignas(64) Avail3Matcher matcher_staging;
+};
+
+struct SvcDev {  // device memory
+  u64 dev_seq;   // published seq for non-leader workgroups
+  u32 dev_cmd;   // SVC_CMD_*
+  u32 matcher_epoch;
+  unsigned long long done;
+  DevCtl ctl;
+  alignas(64) u64 hdr[HDR_WORDS];
+  alignas(64) ttable pool[MAX_GATES];
+  alignas(64) Avail3Matcher matcher;
+};
+
+// ~2 ms of leader poll iterations (each ~1-2 us: one PCIe read + s_sleep).
+constexpr int SVC_IDLE_POLLS = 1500;
+
+__global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
+                                                              SvcDev* dev) {
+  __shared__ alignas(16) u64 s_pool[MAX_GATES * PSTR];
+  __shared__ alignas(16) u8 s_bitmap[256 * 256 / 8];
+  __shared__ u8 s_funs[256];
+  __shared__ alignas(16) u64 s_hdr[HDR_WORDS];
+  __shared__ u64 s_seq;
+  __shared__ u32 s_cmd;
+
+  if (threadIdx.x == 0) {
+    __hip_atomic_fetch_add(&mb->alive, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+
+  const bool leader = blockIdx.x == 0;
+  u64 served = 0;
+
+  for (;;) {
+    // ---- wait for a request (or quit) ----
+    if (threadIdx.x == 0) {
+      u32 cmd = SVC_CMD_WORK;
+      u64 seq = served;
+      if (leader) {
+        int idle = 0;
+        for (;;) {
+          u64 rs = __hip_atomic_load(&mb->req_seq, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_SYSTEM);
+          if (rs != served) {
+            seq = rs;
+            break;
+          }
+          if (__hip_atomic_load(&mb->quit_req, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_SYSTEM) != 0) {
+            cmd = SVC_CMD_QUIT;
+            seq = served + 1;
+            break;
+          }
+          if (++idle > SVC_IDLE_POLLS) {
+            // Retire. After this store the kernel never touches the
+            // mailbox again; the host reaps and relaunches on demand.
+            __hip_atomic_store(&mb->svc_state, SVC_RETIRED, __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+            cmd = SVC_CMD_QUIT;
+            seq = served + 1;
+            break;
+          }
+          __builtin_amdgcn_s_sleep(32);
+        }
+      } else {
+        for (;;) {
+          u64 ds = __hip_atomic_load(&dev->dev_seq, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+          if (ds != served) {
+            seq = ds;
+            cmd = __hip_atomic_load(&dev->dev_cmd, __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
+            break;
+          }
+          __builtin_amdgcn_s_sleep(16);
+        }
+      }
+      s_seq = seq;
+      s_cmd = cmd;
+    }
+    __syncthreads();
+    // One acquire per request, by every thread (poll loads are relaxed):
+    // orders all later payload reads (mailbox or device pool/matcher/hdr)
+    // after the observed seq store on every lane.
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+    const u64 seq = s_seq;
+    const u32 cmd = s_cmd;
+
+    if (leader) {
+      if (cmd == SVC_CMD_WORK) {
+        // Pull the request: header block, pool delta, matcher delta.
+        if (threadIdx.x < HDR_WORDS) s_hdr[threadIdx.x] = mb->hdr[threadIdx.x];
+        __syncthreads();
+        const int n = static_cast<int>(s_hdr[HDR_N_KEEP] & 0xFFFFFFFFu);
+        const int keep = static_cast<int>(s_hdr[HDR_N_KEEP] >> 32);
+        const u32 epoch = static_cast<u32>(s_hdr[HDR_EPOCH_CALL] & 0xFFFFFFFFu);
+        if (threadIdx.x < HDR_WORDS) dev->hdr[threadIdx.x] = s_hdr[threadIdx.x];
+        {
+          const u64* src = reinterpret_cast<const u64*>(mb->pool_staging);
+          u64* dst = reinterpret_cast<u64*>(dev->pool);
+          for (int i = keep * 4 + threadIdx.x; i < n * 4; i += blockDim.x) {
+            dst[i] = src[i];
+          }
+        }
+        if (epoch != dev->matcher_epoch) {
+          const u64* ms = reinterpret_cast<const u64*>(&mb->matcher_staging);
+          u64* md = reinterpret_cast<u64*>(&dev->matcher);
+          for (size_t i = threadIdx.x; i < sizeof(Avail3Matcher) / 8;
+               i += blockDim.x) {
+            md[i] = ms[i];
+          }
+          __syncthreads();
+          if (threadIdx.x == 0) dev->matcher_epoch = epoch;
+        }
+        if (threadIdx.x == 0) {
+          dev->done = 0;
+          dev->ctl.abort = 0;
+          dev->ctl.lock = 0;
+          dev->ctl.found = 0;
+          dev->ctl.evaluated = 0;
+          dev->ctl.queue = 0;
+          dev->ctl.hit_count = 0;
+          dev->ctl.overflow = 0;
+        }
+        __syncthreads();
+      }
+      if (threadIdx.x == 0) {
+        __threadfence();  // agent-release the pool/header/ctl writes
+        __hip_atomic_store(&dev->dev_cmd, cmd, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        __hip_atomic_store(&dev->dev_seq, seq, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_AGENT);
+      }
+      __syncthreads();
+    }
+    if (cmd != SVC_CMD_WORK) return;
+    served = seq;
+
+    // ---- stage request + pool + matcher into LDS ----
+    if (!leader) {
+      if (threadIdx.x < HDR_WORDS) s_hdr[threadIdx.x] = dev->hdr[threadIdx.x];
+    }
+    __syncthreads();
+    const int n = static_cast<int>(s_hdr[HDR_N_KEEP] & 0xFFFFFFFFu);
+    const int count_all = static_cast<int>(s_hdr[HDR_EPOCH_CALL] >> 32);
+    const i64 begin = static_cast<i64>(s_hdr[HDR_BEGIN]);
+    const i64 end = static_cast<i64>(s_hdr[HDR_END]);
+    ttable T1, T0;
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      T1.w[w] = s_hdr[HDR_T1 + w];
+      T0.w[w] = s_hdr[HDR_T0 + w];
+    }
+    load_pool_lds(s_pool, dev->pool, n);
+    for (int i = threadIdx.x; i < 256 * 256 / 8 / 8; i += blockDim.x) {
+      reinterpret_cast<u64*>(s_bitmap)[i] =
+          reinterpret_cast<const u64*>(dev->matcher.bitmap)[i];
+    }
+    for (int i = threadIdx.x; i < 256 / 8; i += blockDim.x) {
+      reinterpret_cast<u64*>(s_funs)[i] =
+          reinterpret_cast<const u64*>(dev->matcher.funs)[i];
+    }
+    __shared__ int s_count;
+    __shared__ unsigned long long s_eval;
+    if (threadIdx.x == 0) {
+      s_count = dev->matcher.count;
+      s_eval = 0;
+    }
+    __syncthreads();
+
+    // ---- scan ----
+    const i64 stride = static_cast<i64>(gridDim.x) * blockDim.x;
+    const i64 idx0 =
+        begin + blockIdx.x * static_cast<i64>(blockDim.x) + threadIdx.x;
+    u64 local_eval = scan4_loop(s_pool, s_bitmap, s_funs, s_count, n, T1, T0,
+                                count_all, &dev->ctl, idx0, end, stride);
+
+    // ---- completion: last workgroup publishes the response ----
+    atomicAdd(&s_eval, static_cast<unsigned long long>(local_eval));
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      if (s_eval != 0) atomicAdd(&dev->ctl.evaluated, s_eval);
+      __threadfence();
+      unsigned long long d = __hip_atomic_fetch_add(
+          &dev->done, 1ULL, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+      if (d == gridDim.x - 1ULL) {
+        mb->r_evaluated = dev->ctl.evaluated;
+        mb->r_found = dev->ctl.found;
+        for (int i = 0; i < 10; i++) mb->r_res[i] = dev->ctl.res[i];
+        __threadfence_system();
+        __hip_atomic_store(&mb->resp_seq, seq, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+      }
+    }
+  }
+}
+
 }  // namespace
+
+// ---------------------------------------------------------------------------
+// ScanService — host side of the persistent k=4 scan kernel. One instance
+// per device per process (gate-mode engines and --jobs workers share it;
+// requests serialize on a mutex, which is correct because every request is
+// a whole-device scan anyway).
+// ---------------------------------------------------------------------------
+class ScanService {
+ public:
+  static std::shared_ptr<ScanService> acquire(int device, std::string* err);
+  ~ScanService();
+
+  // Blocking k=4 scan via the resident kernel. Thread-safe.
+  ScanResult scan4(const ScanRequest& rq, i64 begin, i64 end);
+
+  // Retire the resident kernel if it is running (used before launching
+  // other kernels so an idle service never delays them). Cheap when the
+  // kernel already retired itself.
+  void park();
+
+ private:
+  explicit ScanService(int device);
+  void ensure_running_locked();
+  void quit_locked();
+
+  std::mutex mu_;
+  int device_;
+  hipStream_t stream_ = nullptr;      // service kernel lives here
+  hipStream_t esc_stream_ = nullptr;  // escape-hatch async writes
+  SvcMailbox* mb_ = nullptr;          // pinned fine-grained
+  SvcDev* d_svc_ = nullptr;
+  u64 seq_ = 0;
+  int grid_ = 0;
+  bool running_ = false;
+  int shadow_n_ = 0;                  // valid prefix of mb_->pool_staging
+  u32 matcher_epoch_ = 0;
+};
+
+namespace {
+std::mutex g_svc_mu;
+std::map<int, std::weak_ptr<ScanService>> g_svc_registry;
+
+// Size of the SvcDev control header (everything the host must reset before
+// a relaunch; pool/matcher content survives, matcher_epoch reset forces a
+// re-copy).
+constexpr size_t SVC_DEV_CTL_BYTES = offsetof(SvcDev, hdr);
+}  // namespace
+
+ScanService::ScanService(int device) : device_(device) {
+  SBG_HIP_CHECK(hipSetDevice(device_));
+  SBG_HIP_CHECK(hipStreamCreate(&stream_));
+  SBG_HIP_CHECK(hipStreamCreate(&esc_stream_));
+  SBG_HIP_CHECK(hipHostMalloc(&mb_, sizeof(SvcMailbox), hipHostMallocCoherent));
+  std::memset(mb_, 0, sizeof(SvcMailbox));
+  SBG_HIP_CHECK(hipMalloc(&d_svc_, sizeof(SvcDev)));
+  SBG_HIP_CHECK(hipMemset(d_svc_, 0, sizeof(SvcDev)));
+  int per_cu = 0;
+  SBG_HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+      &per_cu, reinterpret_cast<const void*>(k_scan4_service), SCAN_BLOCK, 0));
+  hipDeviceProp_t prop;
+  SBG_HIP_CHECK(hipGetDeviceProperties(&prop, device_));
+  grid_ = std::clamp(per_cu * prop.multiProcessorCount, 8, 2048);
+}
+
+ScanService::~ScanService() {
+  std::lock_guard<std::mutex> lk(mu_);
+  bool freed_ok = true;
+  if (running_) {
+    (void)hipSetDevice(device_);
+    __atomic_store_n(&mb_->quit_req, 1u, __ATOMIC_RELEASE);
+    // Belt and braces: wake the non-leader workgroups and abort any scan
+    // even if the leader workgroup is wedged.
+    static const u64 big_seq = ~0ULL;
+    static const u32 one = 1;
+    (void)hipMemcpyAsync(&d_svc_->dev_cmd, &one, sizeof(u32),
+                         hipMemcpyHostToDevice, esc_stream_);
+    (void)hipMemcpyAsync(&d_svc_->ctl.abort, &one, sizeof(u32),
+                         hipMemcpyHostToDevice, esc_stream_);
+    (void)hipMemcpyAsync(&d_svc_->dev_seq, &big_seq, sizeof(u64),
+                         hipMemcpyHostToDevice, esc_stream_);
+    (void)hipStreamSynchronize(esc_stream_);
+    const auto t0 = std::chrono::steady_clock::now();
+    for (;;) {
+      hipError_t q = hipStreamQuery(stream_);
+      if (q != hipErrorNotReady) break;
+      if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(5)) {
+        std::fprintf(stderr,
+                     "sboxgates: scan service kernel did not exit; leaking "
+                     "its resources\n");
+        freed_ok = false;
+        break;
+      }
+      std::this_thread::yield();
+    }
+  }
+  if (freed_ok) {
+    if (d_svc_ != nullptr) (void)hipFree(d_svc_);
+    if (mb_ != nullptr) (void)hipHostFree(mb_);
+    if (stream_ != nullptr) (void)hipStreamDestroy(stream_);
+    if (esc_stream_ != nullptr) (void)hipStreamDestroy(esc_stream_);
+  }
+}
+
+std::shared_ptr<ScanService> ScanService::acquire(int device, std::string* err) {
+  std::lock_guard<std::mutex> lk(g_svc_mu);
+  auto& slot = g_svc_registry[device];
+  auto p = slot.lock();
+  if (p != nullptr) return p;
+  try {
+    p = std::shared_ptr<ScanService>(new ScanService(device));
+  } catch (const std::exception& e) {
+    if (err != nullptr) *err = e.what();
+    return nullptr;
+  }
+  slot = p;
+  return p;
+}
+
+void ScanService::ensure_running_locked() {
+  if (running_) {
+    if (__atomic_load_n(&mb_->svc_state, __ATOMIC_ACQUIRE) == SVC_RUNNING) {
+      return;
+    }
+    // The kernel retired itself; reap it.
+    SBG_HIP_CHECK(hipStreamSynchronize(stream_));
+    running_ = false;
+  }
+  SBG_HIP_CHECK(hipMemset(d_svc_, 0, SVC_DEV_CTL_BYTES));
+  mb_->quit_req = 0;
+  mb_->alive = 0;
+  __atomic_store_n(&mb_->svc_state, SVC_RUNNING, __ATOMIC_RELEASE);
+  hipLaunchKernelGGL(k_scan4_service, dim3(grid_), dim3(SCAN_BLOCK), 0, stream_,
+                     mb_, d_svc_);
+  SBG_HIP_CHECK(hipGetLastError());
+  running_ = true;
+  // Residency self-check: all workgroups must report in, otherwise the
+  // completion protocol would deadlock. (Occupancy-API sizing should make
+  // this impossible; a failure means something else holds CUs.)
+  const auto t0 = std::chrono::steady_clock::now();
+  while (__atomic_load_n(&mb_->alive, __ATOMIC_ACQUIRE) <
+         static_cast<u32>(grid_)) {
+    if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(2)) {
+      quit_locked();
+      throw std::runtime_error("scan service grid failed to become resident");
+    }
+    std::this_thread::yield();
+  }
+}
+
+void ScanService::quit_locked() {
+  if (!running_) return;
+  __atomic_store_n(&mb_->quit_req, 1u, __ATOMIC_RELEASE);
+  static const u64 big_seq = ~0ULL;
+  static const u32 one = 1;
+  SBG_HIP_CHECK(hipMemcpyAsync(&d_svc_->dev_cmd, &one, sizeof(u32),
+                               hipMemcpyHostToDevice, esc_stream_));
+  SBG_HIP_CHECK(hipMemcpyAsync(&d_svc_->ctl.abort, &one, sizeof(u32),
+                               hipMemcpyHostToDevice, esc_stream_));
+  SBG_HIP_CHECK(hipMemcpyAsync(&d_svc_->dev_seq, &big_seq, sizeof(u64),
+                               hipMemcpyHostToDevice, esc_stream_));
+  SBG_HIP_CHECK(hipStreamSynchronize(esc_stream_));
+  SBG_HIP_CHECK(hipStreamSynchronize(stream_));
+  running_ = false;
+  mb_->quit_req = 0;
+}
+
+void ScanService::park() {
+  std::lock_guard<std::mutex> lk(mu_);
+  if (!running_) return;
+  (void)hipSetDevice(device_);
+  if (__atomic_load_n(&mb_->svc_state, __ATOMIC_ACQUIRE) == SVC_RETIRED) {
+    SBG_HIP_CHECK(hipStreamSynchronize(stream_));
+    running_ = false;
+    return;
+  }
+  quit_locked();
+}
+
+ScanResult ScanService::scan4(const ScanRequest& rq, i64 begin, i64 end) {
+  std::lock_guard<std::mutex> lk(mu_);
+  SBG_HIP_CHECK(hipSetDevice(device_));
+
+  // Matcher delta (content compare: pointers can be reused across engine
+  // lifetimes, so identity alone is not a safe key).
+  if (std::memcmp(&mb_->matcher_staging, rq.matcher, sizeof(Avail3Matcher)) !=
+      0) {
+    std::memcpy(&mb_->matcher_staging, rq.matcher, sizeof(Avail3Matcher));
+    matcher_epoch_ += 1;
+  }
+
+  // Pool delta against the pinned shadow copy.
+  int keep = 0;
+  {
+    const u64* a = reinterpret_cast<const u64*>(mb_->pool_staging);
+    const u64* b = reinterpret_cast<const u64*>(rq.tables);
+    const int limit = std::min(shadow_n_, rq.n) * 4;
+    int i = 0;
+    while (i < limit && a[i] == b[i]) i++;
+    keep = i / 4;
+    if (keep < rq.n) {
+      std::memcpy(mb_->pool_staging + keep, rq.tables + keep,
+                  sizeof(ttable) * static_cast<size_t>(rq.n - keep));
+    }
+    shadow_n_ = rq.n;
+  }
+
+  // Publish the request.
+  const ttable T1 = rq.target & rq.mask;
+  const ttable T0 = ~rq.target & rq.mask;
+  mb_->hdr[HDR_N_KEEP] = static_cast<u64>(static_cast<u32>(rq.n)) |
+                         (static_cast<u64>(static_cast<u32>(keep)) << 32);
+  mb_->hdr[HDR_EPOCH_CALL] =
+      static_cast<u64>(matcher_epoch_) |
+      (static_cast<u64>(rq.count_all ? 1u : 0u) << 32);
+  mb_->hdr[HDR_BEGIN] = static_cast<u64>(begin);
+  mb_->hdr[HDR_END] = static_cast<u64>(end);
+  mb_->hdr[HDR_SEED] = rq.seed;
+  for (int w = 0; w < 4; w++) {
+    mb_->hdr[HDR_T1 + w] = T1.w[w];
+    mb_->hdr[HDR_T0 + w] = T0.w[w];
+  }
+  ensure_running_locked();
+  const u64 s = ++seq_;
+  __atomic_store_n(&mb_->req_seq, s, __ATOMIC_RELEASE);
+
+  // Wait for the response. Every exit path is bounded: a retire race
+  // relaunches, a stuck scan is aborted via the escape stream.
+  const auto t0 = std::chrono::steady_clock::now();
+  bool escalated = false;
+  int spins = 0;
+  for (;;) {
+    if (__atomic_load_n(&mb_->resp_seq, __ATOMIC_ACQUIRE) == s) break;
+    if (__atomic_load_n(&mb_->svc_state, __ATOMIC_ACQUIRE) == SVC_RETIRED) {
+      // The kernel retired just as we published: reap and relaunch; the
+      // request is still in the mailbox and the fresh kernel serves it.
+      SBG_HIP_CHECK(hipStreamSynchronize(stream_));
+      running_ = false;
+      ensure_running_locked();
+    }
+    const auto dt = std::chrono::steady_clock::now() - t0;
+    if (!escalated && dt > std::chrono::seconds(15)) {
+      escalated = true;
+      static const u32 one = 1;
+      (void)hipMemcpyAsync(&d_svc_->ctl.abort, &one, sizeof(u32),
+                           hipMemcpyHostToDevice, esc_stream_);
+      (void)hipStreamSynchronize(esc_stream_);
+    }
+    if (dt > std::chrono::seconds(30)) {
+      quit_locked();
+      throw std::runtime_error("scan service request timed out");
+    }
+    if (((spins++) & 0x3FF) == 0x3FF) std::this_thread::yield();
+  }
+  if (escalated) {
+    quit_locked();
+    throw std::runtime_error("scan service request needed an abort");
+  }
+
+  ScanResult out;
+  out.evaluated = mb_->r_evaluated;
+  if (mb_->r_found != 0) {
+    out.found = true;
+    std::memcpy(out.res, mb_->r_res, sizeof(out.res));
+  }
+  return out;
+}
 
 // ---------------------------------------------------------------------------
 // GpuEngine host wrapper.
@@ -935,6 +1481,9 @@ struct GpuEngine::Impl {
   u64 hit_cap = 0;
   DevCtl* h_ctl = nullptr;  // pinned staging
   std::string name;
+  std::shared_ptr<ScanService> svc;  // lazily acquired for k=4 scans
+  bool svc_tried = false;
+  bool svc_broken = false;
 
   ~Impl() {
     if (d_pool != nullptr) (void)hipFree(d_pool);
@@ -999,6 +1548,25 @@ GpuEngine::~GpuEngine() { delete impl_; }
 int GpuEngine::device() const { return impl_->device; }
 std::string GpuEngine::device_name() const { return impl_->name; }
 
+bool GpuEngine::scan4_service_active() {
+  Impl* im = impl_;
+  if (im->svc != nullptr) return true;
+  if (im->svc_broken || im->svc_tried) return false;
+  im->svc_tried = true;
+  const char* off = std::getenv("SBOXGATES_NO_SVC");
+  if (off != nullptr && off[0] != '\0' && off[0] != '0') {
+    im->svc_broken = true;
+    return false;
+  }
+  std::string err;
+  im->svc = ScanService::acquire(im->device, &err);
+  if (im->svc == nullptr) {
+    im->svc_broken = true;
+    return false;
+  }
+  return true;
+}
+
 ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   ScanResult out;
   const i64 total = n_choose_k(rq.n, k == 4 ? 3 : k);
@@ -1008,6 +1576,22 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
 
   Impl* im = impl_;
   SBG_HIP_CHECK(hipSetDevice(im->device));
+
+  if (k == 4 && scan4_service_active()) {
+    if (rq.matcher == nullptr) throw std::runtime_error("scan4 needs matcher");
+    try {
+      return im->svc->scan4(rq, begin, end);
+    } catch (const std::exception& e) {
+      // Service failure (residency/timeout): permanently fall back to the
+      // one-shot launch path for this engine.
+      std::fprintf(stderr, "sboxgates: scan service disabled: %s\n", e.what());
+      im->svc.reset();
+      im->svc_broken = true;
+    }
+  } else if (k != 4 && im->svc != nullptr) {
+    // Do not let an idle-resident service delay other kernels.
+    im->svc->park();
+  }
 
   // Upload the pool (through pinned staging) and reset the control block.
   std::memcpy(im->h_pool, rq.tables, sizeof(ttable) * rq.n);

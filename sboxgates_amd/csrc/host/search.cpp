@@ -272,8 +272,20 @@ ScanResult Engine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   // the range size where the GPU (incl. ~20-40us launch+upload overhead)
   // beats the CPU differs by ~100x across kinds. Measured on the AES bit-0
   // search: a size-only 2^16 threshold left 2.1e9 7-LUT combos on the CPU
-  // and tripled the wall time.
-  const i64 gpu_min = k == 7 ? 256 : (k == 5 ? 4096 : 1 << 14);
+  // and tripled the wall time. k=4 scans go through the persistent scan
+  // service (~5-8 us/call vs the ~30 us one-shot floor), which moves the
+  // cutover well down.
+  static const i64 min4_svc = [] {
+    const char* s = std::getenv("SBOXGATES_GPU_MIN4");
+    return s != nullptr ? std::strtoll(s, nullptr, 10) : 4096;
+  }();
+  const i64 gpu_min =
+      k == 7 ? 256
+             : (k == 5 ? 4096
+                       : (k == 4 && gpu_ != nullptr &&
+                                  gpu_->scan4_service_active()
+                              ? min4_svc
+                              : 1 << 14));
   bool use_gpu = gpu_ != nullptr && (opt_.gpu == GPU_FORCE || end - begin >= gpu_min);
   const auto t0 = std::chrono::steady_clock::now();
   ScanResult r;
@@ -674,7 +686,7 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
     gatenum nst_out = NO_GATE;
 
     if (opt_.lut_graph) {  // LUT-based multiplexer.
-      nst = *st;
+      copy_state(nst, *st);
       nst.max_gates -= 1;  // Room for the multiplexer.
       gatenum fb = create_circuit(&nst, target, mask & ~fsel, next_inbits);
       if (fb == NO_GATE) continue;
@@ -698,7 +710,8 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
       }
       assert(tt_eq_mask(target, nst.gates[nst_out].table, mask));
     } else {  // Try both AND- and OR-based multiplexers; keep the smaller.
-      state nst_and = *st;
+      state nst_and;
+      copy_state(nst_and, *st);
       nst_and.max_gates -= 2;
       nst_and.max_sat_metric -= sat_metric_of(AND) + sat_metric_of(XOR);
 
@@ -713,7 +726,8 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
         mux_out_and = add_xor_gate(&nst_and, fb, andg, opt_.metric);
       }
 
-      state nst_or = *st;
+      state nst_or;
+      copy_state(nst_or, *st);
       if (mux_out_and != NO_GATE) {
         nst_or.max_gates = nst_and.num_gates;
         nst_or.max_sat_metric = nst_and.sat_metric;
@@ -744,10 +758,10 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
                    (mux_out_and != NO_GATE && nst_and.sat_metric < nst_or.sat_metric);
       }
       if (pick_and) {
-        nst = nst_and;
+        copy_state(nst, nst_and);
         nst_out = mux_out_and;
       } else {
-        nst = nst_or;
+        copy_state(nst, nst_or);
         nst_out = mux_out_or;
       }
     }
@@ -755,19 +769,19 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
     // Keep the best sub-state by the active metric (sboxgates.c:593-606).
     if (opt_.metric == METRIC_GATES) {
       if (best.num_gates == 0 || nst.num_gates < best.num_gates) {
-        best = nst;
+        copy_state(best, nst);
         best_out = nst_out;
       }
     } else {
       if (best.sat_metric == 0 || nst.sat_metric < best.sat_metric) {
-        best = nst;
+        copy_state(best, nst);
         best_out = nst_out;
       }
     }
   }
 
   if (best.num_gates == 0) return NO_GATE;
-  *st = best;
+  copy_state(*st, best);
   return assert_ret(best_out, target, st, mask, "step5");
 }
 

@@ -19,9 +19,14 @@ class GpuEngine {
   static std::unique_ptr<GpuEngine> create(int device, std::string* err);
   ~GpuEngine();
 
-  // Scans combinations [begin, end) of C(rq.n, k), k in {3,5,7}.
+  // Scans combinations [begin, end) of C(rq.n, k), k in {3,4,5,7}.
   // Blocking; early-exits via an in-kernel abort flag unless rq.count_all.
   ScanResult scan(int k, const ScanRequest& rq, i64 begin, i64 end);
+
+  // True when k=4 scans go through the persistent scan-service kernel
+  // (~5-8 us per call) instead of one-shot launches (~30 us API floor).
+  // Lazily creates the per-device service on first query.
+  bool scan4_service_active();
 
   int device() const;
   std::string device_name() const;

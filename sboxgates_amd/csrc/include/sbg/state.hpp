@@ -10,6 +10,8 @@
 #pragma once
 
 #include <climits>
+#include <cstddef>
+#include <cstring>
 #include <string>
 
 #include "sbg/boolfunc.hpp"
@@ -40,6 +42,16 @@ struct state {
 };
 static_assert(sizeof(state) == 32 + 64 * MAX_GATES,
               "state layout must match the reference");
+
+// Copies only the live prefix (header + gates[0..num_gates)) — the struct
+// is 32 KB but live circuits are usually < 100 gates, and the step-5 mux
+// recursion copies states several times per node. Dead slots in dst are
+// left stale, which is safe: every consumer (fingerprint, XML writer, DAG
+// eval, scans, rank broadcast) interprets only the live prefix.
+inline void copy_state(state& dst, const state& src) {
+  std::memcpy(&dst, &src,
+              offsetof(state, gates) + sizeof(gate) * src.num_gates);
+}
 
 // SAT-metric cost model (parity: state.c:168-191). Calling with LUT aborts
 // in the reference; here it returns a sentinel the callers treat as invalid.

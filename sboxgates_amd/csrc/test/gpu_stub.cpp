@@ -19,5 +19,6 @@ GpuEngine::~GpuEngine() = default;
 int GpuEngine::device() const { return -1; }
 std::string GpuEngine::device_name() const { return "stub"; }
 ScanResult GpuEngine::scan(int, const ScanRequest&, i64, i64) { return {}; }
+bool GpuEngine::scan4_service_active() { return false; }
 
 }  // namespace sbg
