@@ -1221,6 +1221,7 @@ class ScanService {
   explicit ScanService(int device);
   void ensure_running_locked();
   void quit_locked();
+  bool quit_bounded();
 
   std::mutex mu_;
   int device_;
@@ -1229,6 +1230,7 @@ class ScanService {
   SvcMailbox* mb_ = nullptr;          // pinned fine-grained
   SvcDev* d_svc_ = nullptr;
   u64 seq_ = 0;
+  u64 relaunches_ = 0;
   int grid_ = 0;
   bool running_ = false;
   int shadow_n_ = 0;                  // valid prefix of mb_->pool_staging
@@ -1245,6 +1247,14 @@ std::map<int, std::weak_ptr<ScanService>> g_svc_registry;
 constexpr size_t SVC_DEV_CTL_BYTES = offsetof(SvcDev, hdr);
 }  // namespace
 
+static bool svc_debug() {
+  static const bool on = [] {
+    const char* s = std::getenv("SBOXGATES_SVC_DEBUG");
+    return s != nullptr && s[0] != '\0' && s[0] != '0';
+  }();
+  return on;
+}
+
 ScanService::ScanService(int device) : device_(device) {
   SBG_HIP_CHECK(hipSetDevice(device_));
   SBG_HIP_CHECK(hipStreamCreate(&stream_));
@@ -1259,6 +1269,14 @@ ScanService::ScanService(int device) : device_(device) {
   hipDeviceProp_t prop;
   SBG_HIP_CHECK(hipGetDeviceProperties(&prop, device_));
   grid_ = std::clamp(per_cu * prop.multiProcessorCount, 8, 2048);
+  if (const char* g = std::getenv("SBOXGATES_SVC_GRID")) {
+    long v = std::strtol(g, nullptr, 10);
+    if (v >= 1 && v <= 4096) grid_ = static_cast<int>(v);
+  }
+  if (svc_debug()) {
+    std::fprintf(stderr, "[svc] device %d: occupancy %d/CU x %d CUs -> grid %d\n",
+                 device_, per_cu, prop.multiProcessorCount, grid_);
+  }
 }
 
 ScanService::~ScanService() {
@@ -1266,30 +1284,11 @@ ScanService::~ScanService() {
   bool freed_ok = true;
   if (running_) {
     (void)hipSetDevice(device_);
-    __atomic_store_n(&mb_->quit_req, 1u, __ATOMIC_RELEASE);
-    // Belt and braces: wake the non-leader workgroups and abort any scan
-    // even if the leader workgroup is wedged.
-    static const u64 big_seq = ~0ULL;
-    static const u32 one = 1;
-    (void)hipMemcpyAsync(&d_svc_->dev_cmd, &one, sizeof(u32),
-                         hipMemcpyHostToDevice, esc_stream_);
-    (void)hipMemcpyAsync(&d_svc_->ctl.abort, &one, sizeof(u32),
-                         hipMemcpyHostToDevice, esc_stream_);
-    (void)hipMemcpyAsync(&d_svc_->dev_seq, &big_seq, sizeof(u64),
-                         hipMemcpyHostToDevice, esc_stream_);
-    (void)hipStreamSynchronize(esc_stream_);
-    const auto t0 = std::chrono::steady_clock::now();
-    for (;;) {
-      hipError_t q = hipStreamQuery(stream_);
-      if (q != hipErrorNotReady) break;
-      if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(5)) {
-        std::fprintf(stderr,
-                     "sboxgates: scan service kernel did not exit; leaking "
-                     "its resources\n");
-        freed_ok = false;
-        break;
-      }
-      std::this_thread::yield();
+    freed_ok = quit_bounded();
+    if (!freed_ok) {
+      std::fprintf(stderr,
+                   "sboxgates: scan service kernel did not exit; leaking its "
+                   "resources\n");
     }
   }
   if (freed_ok) {
@@ -1297,6 +1296,41 @@ ScanService::~ScanService() {
     if (mb_ != nullptr) (void)hipHostFree(mb_);
     if (stream_ != nullptr) (void)hipStreamDestroy(stream_);
     if (esc_stream_ != nullptr) (void)hipStreamDestroy(esc_stream_);
+  }
+}
+
+// Ask the resident kernel to exit and wait for it, bounded. The primary
+// channel is the pinned quit_req word (a plain host store — always lands):
+// the leader workgroup polls it between requests and broadcasts QUIT to
+// the other workgroups through device stores (which, being on-device,
+// cannot be blocked by anything). The hipMemcpy escalation is best-effort
+// only: small H2D copies may be executed as blit kernels, which cannot get
+// CUs while the persistent grid holds them.
+bool ScanService::quit_bounded() {
+  __atomic_store_n(&mb_->quit_req, 1u, __ATOMIC_RELEASE);
+  const auto t0 = std::chrono::steady_clock::now();
+  bool escalated = false;
+  for (;;) {
+    hipError_t q = hipStreamQuery(stream_);
+    if (q != hipErrorNotReady) {
+      running_ = false;
+      mb_->quit_req = 0;
+      return true;
+    }
+    const auto dt = std::chrono::steady_clock::now() - t0;
+    if (!escalated && dt > std::chrono::seconds(5)) {
+      escalated = true;
+      static const u32 one = 1;
+      static const u64 big_seq = ~0ULL;
+      (void)hipMemcpyAsync(&d_svc_->ctl.abort, &one, sizeof(u32),
+                           hipMemcpyHostToDevice, esc_stream_);
+      (void)hipMemcpyAsync(&d_svc_->dev_cmd, &one, sizeof(u32),
+                           hipMemcpyHostToDevice, esc_stream_);
+      (void)hipMemcpyAsync(&d_svc_->dev_seq, &big_seq, sizeof(u64),
+                           hipMemcpyHostToDevice, esc_stream_);
+    }
+    if (dt > std::chrono::seconds(10)) return false;
+    std::this_thread::yield();
   }
 }
 
@@ -1332,6 +1366,7 @@ void ScanService::ensure_running_locked() {
                      mb_, d_svc_);
   SBG_HIP_CHECK(hipGetLastError());
   running_ = true;
+  relaunches_ += 1;
   // Residency self-check: all workgroups must report in, otherwise the
   // completion protocol would deadlock. (Occupancy-API sizing should make
   // this impossible; a failure means something else holds CUs.)
@@ -1339,28 +1374,30 @@ void ScanService::ensure_running_locked() {
   while (__atomic_load_n(&mb_->alive, __ATOMIC_ACQUIRE) <
          static_cast<u32>(grid_)) {
     if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(2)) {
+      const u32 alive = __atomic_load_n(&mb_->alive, __ATOMIC_ACQUIRE);
+      if (svc_debug()) {
+        std::fprintf(stderr, "[svc] residency failure: %u of %d alive\n", alive,
+                     grid_);
+      }
       quit_locked();
       throw std::runtime_error("scan service grid failed to become resident");
     }
     std::this_thread::yield();
   }
+  if (svc_debug()) {
+    std::fprintf(stderr, "[svc] launch #%llu: %d workgroups resident in %.1f us\n",
+                 static_cast<unsigned long long>(relaunches_), grid_,
+                 std::chrono::duration<double, std::micro>(
+                     std::chrono::steady_clock::now() - t0)
+                     .count());
+  }
 }
 
 void ScanService::quit_locked() {
   if (!running_) return;
-  __atomic_store_n(&mb_->quit_req, 1u, __ATOMIC_RELEASE);
-  static const u64 big_seq = ~0ULL;
-  static const u32 one = 1;
-  SBG_HIP_CHECK(hipMemcpyAsync(&d_svc_->dev_cmd, &one, sizeof(u32),
-                               hipMemcpyHostToDevice, esc_stream_));
-  SBG_HIP_CHECK(hipMemcpyAsync(&d_svc_->ctl.abort, &one, sizeof(u32),
-                               hipMemcpyHostToDevice, esc_stream_));
-  SBG_HIP_CHECK(hipMemcpyAsync(&d_svc_->dev_seq, &big_seq, sizeof(u64),
-                               hipMemcpyHostToDevice, esc_stream_));
-  SBG_HIP_CHECK(hipStreamSynchronize(esc_stream_));
-  SBG_HIP_CHECK(hipStreamSynchronize(stream_));
-  running_ = false;
-  mb_->quit_req = 0;
+  if (!quit_bounded()) {
+    throw std::runtime_error("scan service kernel refused to exit");
+  }
 }
 
 void ScanService::park() {
@@ -1426,6 +1463,7 @@ ScanResult ScanService::scan4(const ScanRequest& rq, i64 begin, i64 end) {
   // relaunches, a stuck scan is aborted via the escape stream.
   const auto t0 = std::chrono::steady_clock::now();
   bool escalated = false;
+  bool reported = false;
   int spins = 0;
   for (;;) {
     if (__atomic_load_n(&mb_->resp_seq, __ATOMIC_ACQUIRE) == s) break;
@@ -1437,14 +1475,23 @@ ScanResult ScanService::scan4(const ScanRequest& rq, i64 begin, i64 end) {
       ensure_running_locked();
     }
     const auto dt = std::chrono::steady_clock::now() - t0;
-    if (!escalated && dt > std::chrono::seconds(15)) {
+    if (svc_debug() && !reported && dt > std::chrono::seconds(1)) {
+      reported = true;
+      std::fprintf(stderr,
+                   "[svc] slow request: seq=%llu resp=%llu state=%u alive=%u "
+                   "n=%d range=%lld\n",
+                   static_cast<unsigned long long>(s),
+                   static_cast<unsigned long long>(mb_->resp_seq),
+                   mb_->svc_state, mb_->alive, rq.n,
+                   static_cast<long long>(end - begin));
+    }
+    if (!escalated && dt > std::chrono::seconds(10)) {
       escalated = true;
       static const u32 one = 1;
       (void)hipMemcpyAsync(&d_svc_->ctl.abort, &one, sizeof(u32),
                            hipMemcpyHostToDevice, esc_stream_);
-      (void)hipStreamSynchronize(esc_stream_);
     }
-    if (dt > std::chrono::seconds(30)) {
+    if (dt > std::chrono::seconds(20)) {
       quit_locked();
       throw std::runtime_error("scan service request timed out");
     }
