@@ -995,6 +995,11 @@ struct SvcMailbox {  // pinned fine-grained host memory
   u32 alive;         // residency self-check counter
   u16 r_res[10];
   u16 pad[2];
+  // debug trace (SBOXGATES_SVC_DEBUG): per-WG protocol stage + leader poll
+  // counter, plain stores to host memory; read by the host on slow requests.
+  u64 leader_polls;
+  u64 leader_seen_rs;
+  alignas(64) u8 stage[4096];
   // request payload
   alignas(64) u64 hdr[HDR_WORDS];
   alignas(64) ttable pool_staging[MAX_GATES];  // full pool image (host shadow)
@@ -1027,6 +1032,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
   if (threadIdx.x == 0) {
     __hip_atomic_fetch_add(&mb->alive, 1u, __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_SYSTEM);
+    if (blockIdx.x < 4096) mb->stage[blockIdx.x] = 1;  // alive
   }
 
   const bool leader = blockIdx.x == 0;
@@ -1042,6 +1048,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
         for (;;) {
           u64 rs = __hip_atomic_load(&mb->req_seq, __ATOMIC_RELAXED,
                                      __HIP_MEMORY_SCOPE_SYSTEM);
+          mb->leader_polls += 1;
+          mb->leader_seen_rs = rs;
           if (rs != served) {
             seq = rs;
             break;
@@ -1078,6 +1086,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
       }
       s_seq = seq;
       s_cmd = cmd;
+      if (blockIdx.x < 4096) mb->stage[blockIdx.x] = 2;  // got request
     }
     __syncthreads();
     // One acquire per request, by every thread (poll loads are relaxed):
@@ -1131,6 +1140,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
                            __HIP_MEMORY_SCOPE_AGENT);
         __hip_atomic_store(&dev->dev_seq, seq, __ATOMIC_RELEASE,
                            __HIP_MEMORY_SCOPE_AGENT);
+        mb->stage[0] = 3;  // leader published
       }
       __syncthreads();
     }
@@ -1169,6 +1179,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
     }
     __syncthreads();
 
+    if (threadIdx.x == 0 && blockIdx.x < 4096) mb->stage[blockIdx.x] = 4;  // staged
+
     // ---- scan ----
     const i64 stride = static_cast<i64>(gridDim.x) * blockDim.x;
     const i64 idx0 =
@@ -1180,6 +1192,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
     atomicAdd(&s_eval, static_cast<unsigned long long>(local_eval));
     __syncthreads();
     if (threadIdx.x == 0) {
+      if (blockIdx.x < 4096) mb->stage[blockIdx.x] = 5;  // scanned
       if (s_eval != 0) atomicAdd(&dev->ctl.evaluated, s_eval);
       __threadfence();
       unsigned long long d = __hip_atomic_fetch_add(
@@ -1477,13 +1490,21 @@ ScanResult ScanService::scan4(const ScanRequest& rq, i64 begin, i64 end) {
     const auto dt = std::chrono::steady_clock::now() - t0;
     if (svc_debug() && !reported && dt > std::chrono::seconds(1)) {
       reported = true;
+      int hist[8] = {};
+      for (int i = 0; i < grid_ && i < 4096; i++) {
+        hist[mb_->stage[i] & 7] += 1;
+      }
       std::fprintf(stderr,
                    "[svc] slow request: seq=%llu resp=%llu state=%u alive=%u "
-                   "n=%d range=%lld\n",
+                   "n=%d range=%lld polls=%llu seen_rs=%llu stages "
+                   "[%d %d %d %d %d %d]\n",
                    static_cast<unsigned long long>(s),
                    static_cast<unsigned long long>(mb_->resp_seq),
                    mb_->svc_state, mb_->alive, rq.n,
-                   static_cast<long long>(end - begin));
+                   static_cast<long long>(end - begin),
+                   static_cast<unsigned long long>(mb_->leader_polls),
+                   static_cast<unsigned long long>(mb_->leader_seen_rs),
+                   hist[0], hist[1], hist[2], hist[3], hist[4], hist[5]);
     }
     if (!escalated && dt > std::chrono::seconds(10)) {
       escalated = true;
