@@ -1099,8 +1099,10 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
     if (leader) {
       if (cmd == SVC_CMD_WORK) {
         // Pull the request: header block, pool delta, matcher delta.
+        if (threadIdx.x == 0) mb->leader_seen_rs = 101;
         if (threadIdx.x < HDR_WORDS) s_hdr[threadIdx.x] = mb->hdr[threadIdx.x];
         __syncthreads();
+        if (threadIdx.x == 0) mb->leader_seen_rs = 102;
         const int n = static_cast<int>(s_hdr[HDR_N_KEEP] & 0xFFFFFFFFu);
         const int keep = static_cast<int>(s_hdr[HDR_N_KEEP] >> 32);
         const u32 epoch = static_cast<u32>(s_hdr[HDR_EPOCH_CALL] & 0xFFFFFFFFu);
@@ -1112,6 +1114,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
             dst[i] = src[i];
           }
         }
+        if (threadIdx.x == 0) mb->leader_seen_rs = 103;
         if (epoch != dev->matcher_epoch) {
           const u64* ms = reinterpret_cast<const u64*>(&mb->matcher_staging);
           u64* md = reinterpret_cast<u64*>(&dev->matcher);
@@ -1123,6 +1126,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
           if (threadIdx.x == 0) dev->matcher_epoch = epoch;
         }
         if (threadIdx.x == 0) {
+          mb->leader_seen_rs = 104;
           dev->done = 0;
           dev->ctl.abort = 0;
           dev->ctl.lock = 0;
@@ -1133,6 +1137,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
           dev->ctl.overflow = 0;
         }
         __syncthreads();
+        if (threadIdx.x == 0) mb->leader_seen_rs = 105;
       }
       if (threadIdx.x == 0) {
         __threadfence();  // agent-release the pool/header/ctl writes
@@ -1141,6 +1146,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
         __hip_atomic_store(&dev->dev_seq, seq, __ATOMIC_RELEASE,
                            __HIP_MEMORY_SCOPE_AGENT);
         mb->stage[0] = 3;  // leader published
+        mb->leader_seen_rs = 106;
       }
       __syncthreads();
     }
@@ -1180,6 +1186,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
     __syncthreads();
 
     if (threadIdx.x == 0 && blockIdx.x < 4096) mb->stage[blockIdx.x] = 4;  // staged
+    if (leader && threadIdx.x == 0) mb->leader_seen_rs = 107;
 
     // ---- scan ----
     const i64 stride = static_cast<i64>(gridDim.x) * blockDim.x;
@@ -1193,6 +1200,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
     __syncthreads();
     if (threadIdx.x == 0) {
       if (blockIdx.x < 4096) mb->stage[blockIdx.x] = 5;  // scanned
+      if (leader) mb->leader_seen_rs = 108;
       if (s_eval != 0) atomicAdd(&dev->ctl.evaluated, s_eval);
       __threadfence();
       unsigned long long d = __hip_atomic_fetch_add(
