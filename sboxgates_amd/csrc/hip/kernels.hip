@@ -148,20 +148,40 @@ __device__ __forceinline__ u64 dev_rnd(u64 seed, u64 idx) {
   return hash_mix64(seed ^ (idx * 0x9E3779B97F4A7C15ULL));
 }
 
-__device__ __forceinline__ bool dev_abort(const DevCtl* ctl) {
-  return __hip_atomic_load(&ctl->abort, __ATOMIC_RELAXED,
-                           __HIP_MEMORY_SCOPE_AGENT) != 0;
+// Global-address-space pointer casts for cross-workgroup / host-visible
+// words. Per the CDNA4 inter-workgroup rules (cdna_hip_programming.md
+// Guideline 16): every shared word must be a GLOBAL agent/system-scope
+// access — flat atomics miss the cache-bypass lowering and an XCD's
+// private L2 then serves stale values to its readers forever.
+typedef __attribute__((address_space(1))) unsigned svc_gu32;
+typedef __attribute__((address_space(1))) unsigned long long svc_gu64;
+__device__ __forceinline__ svc_gu32* as_gu32(u32* p) { return (svc_gu32*)p; }
+__device__ __forceinline__ svc_gu64* as_gu64(u64* p) { return (svc_gu64*)p; }
+__device__ __forceinline__ svc_gu64* as_gu64(unsigned long long* p) {
+  return (svc_gu64*)p;
 }
 
-// Publish a winner: single writer wins the lock; the rest skip. Host reads
-// after stream sync, so plain stores suffice for the payload; the abort
-// flag is agent-scope so other workgroups see it promptly.
+__device__ __forceinline__ bool dev_abort(const DevCtl* ctl) {
+  return __hip_atomic_load(as_gu32(const_cast<u32*>(&ctl->abort)),
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) != 0;
+}
+
+// Publish a winner: single writer wins the lock; the rest skip. Payload
+// (res) is plain-stored, then drained and released before the found/abort
+// flag stores (Guideline 16 R1: drain -> release fence -> asm wait ->
+// relaxed agent flag store), so an agent-scope reader that acquires after
+// seeing found != 0 reads a complete payload on any XCD. Host readers via
+// stream-sync + memcpy are ordered regardless.
 __device__ __forceinline__ void dev_publish(DevCtl* ctl, const u16 res[10]) {
   if (atomicCAS(&ctl->lock, 0u, 1u) == 0u) {
     for (int i = 0; i < 10; i++) ctl->res[i] = res[i];
-    __threadfence();
-    ctl->found = 1;
-    __hip_atomic_store(&ctl->abort, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __hip_atomic_store(as_gu32(&ctl->found), 1u, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
+    __hip_atomic_store(as_gu32(&ctl->abort), 1u, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
   }
 }
 
@@ -989,17 +1009,11 @@ struct SvcMailbox {  // pinned fine-grained host memory
   u32 quit_req;
   u32 svc_state;     // SVC_RUNNING / SVC_RETIRED
   // device -> host
-  u64 resp_seq;      // release-stored by the last workgroup
+  u64 resp_seq;      // flag-stored by the last workgroup (after a drain)
   u64 r_evaluated;
   u32 r_found;
   u32 alive;         // residency self-check counter
-  u16 r_res[10];
-  u16 pad[2];
-  // debug trace (SBOXGATES_SVC_DEBUG): per-WG protocol stage + leader poll
-  // counter, plain stores to host memory; read by the host on slow requests.
-  u64 leader_polls;
-  u64 leader_seen_rs;
-  alignas(64) u8 stage[4096];
+  u64 r_res_w[3];    // res[10] packed 4 x u16 per word
   // request payload
   alignas(64) u64 hdr[HDR_WORDS];
   alignas(64) ttable pool_staging[MAX_GATES];  // full pool image (host shadow)
@@ -1026,83 +1040,76 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
   __shared__ alignas(16) u8 s_bitmap[256 * 256 / 8];
   __shared__ u8 s_funs[256];
   __shared__ alignas(16) u64 s_hdr[HDR_WORDS];
-  __shared__ u64 s_seq;
-  __shared__ u32 s_cmd;
+  __shared__ u64 s_word;
+  __shared__ int s_count;
+  __shared__ unsigned long long s_eval;
 
   if (threadIdx.x == 0) {
-    __hip_atomic_fetch_add(&mb->alive, 1u, __ATOMIC_RELAXED,
+    __hip_atomic_fetch_add(as_gu32(&mb->alive), 1u, __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_SYSTEM);
-    if (blockIdx.x < 4096) mb->stage[blockIdx.x] = 1;  // alive
   }
 
   const bool leader = blockIdx.x == 0;
+  // Protocol word: (req_seq << 1) | quit. Monotonic; one word carries both
+  // the sequence and the command, so publish/consume is a single-flag
+  // hand-off (Guideline 16 R2 style).
   u64 served = 0;
 
   for (;;) {
     // ---- wait for a request (or quit) ----
     if (threadIdx.x == 0) {
-      u32 cmd = SVC_CMD_WORK;
-      u64 seq = served;
+      u64 word;
       if (leader) {
         int idle = 0;
         for (;;) {
-          u64 rs = __hip_atomic_load(&mb->req_seq, __ATOMIC_RELAXED,
+          u64 rs = __hip_atomic_load(as_gu64(&mb->req_seq), __ATOMIC_RELAXED,
                                      __HIP_MEMORY_SCOPE_SYSTEM);
-          mb->leader_polls += 1;
-          mb->leader_seen_rs = rs;
-          if (rs != served) {
-            seq = rs;
+          if ((rs << 1) != served) {
+            word = rs << 1;
             break;
           }
-          if (__hip_atomic_load(&mb->quit_req, __ATOMIC_RELAXED,
+          if (__hip_atomic_load(as_gu32(&mb->quit_req), __ATOMIC_RELAXED,
                                 __HIP_MEMORY_SCOPE_SYSTEM) != 0) {
-            cmd = SVC_CMD_QUIT;
-            seq = served + 1;
+            word = served + 3;  // bump seq, set quit bit
             break;
           }
           if (++idle > SVC_IDLE_POLLS) {
             // Retire. After this store the kernel never touches the
             // mailbox again; the host reaps and relaunches on demand.
-            __hip_atomic_store(&mb->svc_state, SVC_RETIRED, __ATOMIC_RELEASE,
-                               __HIP_MEMORY_SCOPE_SYSTEM);
-            cmd = SVC_CMD_QUIT;
-            seq = served + 1;
+            __hip_atomic_store(as_gu32(&mb->svc_state), SVC_RETIRED,
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+            word = served + 3;
             break;
           }
           __builtin_amdgcn_s_sleep(32);
         }
       } else {
         for (;;) {
-          u64 ds = __hip_atomic_load(&dev->dev_seq, __ATOMIC_RELAXED,
+          u64 ds = __hip_atomic_load(as_gu64(&dev->dev_seq), __ATOMIC_RELAXED,
                                      __HIP_MEMORY_SCOPE_AGENT);
           if (ds != served) {
-            seq = ds;
-            cmd = __hip_atomic_load(&dev->dev_cmd, __ATOMIC_RELAXED,
-                                    __HIP_MEMORY_SCOPE_AGENT);
+            word = ds;
             break;
           }
           __builtin_amdgcn_s_sleep(16);
         }
       }
-      s_seq = seq;
-      s_cmd = cmd;
-      if (blockIdx.x < 4096) mb->stage[blockIdx.x] = 2;  // got request
+      // ONE acquire after the poll match (covers the workgroup: a
+      // __syncthreads() follows). System scope: the leader's payload
+      // source is host memory.
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+      s_word = word;
     }
     __syncthreads();
-    // One acquire per request, by every thread (poll loads are relaxed):
-    // orders all later payload reads (mailbox or device pool/matcher/hdr)
-    // after the observed seq store on every lane.
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
-    const u64 seq = s_seq;
-    const u32 cmd = s_cmd;
+    const u64 word = s_word;
+    const bool quit = (word & 1) != 0;
+    const u64 seq = word >> 1;
 
     if (leader) {
-      if (cmd == SVC_CMD_WORK) {
+      if (!quit) {
         // Pull the request: header block, pool delta, matcher delta.
-        if (threadIdx.x == 0) mb->leader_seen_rs = 101;
         if (threadIdx.x < HDR_WORDS) s_hdr[threadIdx.x] = mb->hdr[threadIdx.x];
         __syncthreads();
-        if (threadIdx.x == 0) mb->leader_seen_rs = 102;
         const int n = static_cast<int>(s_hdr[HDR_N_KEEP] & 0xFFFFFFFFu);
         const int keep = static_cast<int>(s_hdr[HDR_N_KEEP] >> 32);
         const u32 epoch = static_cast<u32>(s_hdr[HDR_EPOCH_CALL] & 0xFFFFFFFFu);
@@ -1114,7 +1121,6 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
             dst[i] = src[i];
           }
         }
-        if (threadIdx.x == 0) mb->leader_seen_rs = 103;
         if (epoch != dev->matcher_epoch) {
           const u64* ms = reinterpret_cast<const u64*>(&mb->matcher_staging);
           u64* md = reinterpret_cast<u64*>(&dev->matcher);
@@ -1126,7 +1132,6 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
           if (threadIdx.x == 0) dev->matcher_epoch = epoch;
         }
         if (threadIdx.x == 0) {
-          mb->leader_seen_rs = 104;
           dev->done = 0;
           dev->ctl.abort = 0;
           dev->ctl.lock = 0;
@@ -1136,22 +1141,22 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
           dev->ctl.hit_count = 0;
           dev->ctl.overflow = 0;
         }
-        __syncthreads();
-        if (threadIdx.x == 0) mb->leader_seen_rs = 105;
       }
+      // Publish to the other workgroups (Guideline 16 counter form):
+      // every wave drains its plain stores, then one lane releases and
+      // stores the flag word relaxed.
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
       if (threadIdx.x == 0) {
-        __threadfence();  // agent-release the pool/header/ctl writes
-        __hip_atomic_store(&dev->dev_cmd, cmd, __ATOMIC_RELAXED,
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __hip_atomic_store(as_gu64(&dev->dev_seq), word, __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_AGENT);
-        __hip_atomic_store(&dev->dev_seq, seq, __ATOMIC_RELEASE,
-                           __HIP_MEMORY_SCOPE_AGENT);
-        mb->stage[0] = 3;  // leader published
-        mb->leader_seen_rs = 106;
       }
       __syncthreads();
     }
-    if (cmd != SVC_CMD_WORK) return;
-    served = seq;
+    if (quit) return;
+    served = word;
 
     // ---- stage request + pool + matcher into LDS ----
     if (!leader) {
@@ -1177,16 +1182,11 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
       reinterpret_cast<u64*>(s_funs)[i] =
           reinterpret_cast<const u64*>(dev->matcher.funs)[i];
     }
-    __shared__ int s_count;
-    __shared__ unsigned long long s_eval;
     if (threadIdx.x == 0) {
       s_count = dev->matcher.count;
       s_eval = 0;
     }
     __syncthreads();
-
-    if (threadIdx.x == 0 && blockIdx.x < 4096) mb->stage[blockIdx.x] = 4;  // staged
-    if (leader && threadIdx.x == 0) mb->leader_seen_rs = 107;
 
     // ---- scan ----
     const i64 stride = static_cast<i64>(gridDim.x) * blockDim.x;
@@ -1199,18 +1199,43 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
     atomicAdd(&s_eval, static_cast<unsigned long long>(local_eval));
     __syncthreads();
     if (threadIdx.x == 0) {
-      if (blockIdx.x < 4096) mb->stage[blockIdx.x] = 5;  // scanned
-      if (leader) mb->leader_seen_rs = 108;
-      if (s_eval != 0) atomicAdd(&dev->ctl.evaluated, s_eval);
-      __threadfence();
+      if (s_eval != 0) {
+        __hip_atomic_fetch_add(as_gu64(&dev->ctl.evaluated),
+                               static_cast<unsigned long long>(s_eval),
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      }
+      // Fence-then-ticket order (Guideline 16 split-K recipe).
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       unsigned long long d = __hip_atomic_fetch_add(
-          &dev->done, 1ULL, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+          as_gu64(&dev->done), 1ULL, __ATOMIC_RELAXED,
+          __HIP_MEMORY_SCOPE_AGENT);
       if (d == gridDim.x - 1ULL) {
-        mb->r_evaluated = dev->ctl.evaluated;
-        mb->r_found = dev->ctl.found;
-        for (int i = 0; i < 10; i++) mb->r_res[i] = dev->ctl.res[i];
-        __threadfence_system();
-        __hip_atomic_store(&mb->resp_seq, seq, __ATOMIC_RELEASE,
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        const unsigned long long ev = __hip_atomic_load(
+            as_gu64(&dev->ctl.evaluated), __ATOMIC_RELAXED,
+            __HIP_MEMORY_SCOPE_AGENT);
+        const u32 found = __hip_atomic_load(as_gu32(&dev->ctl.found),
+                                            __ATOMIC_RELAXED,
+                                            __HIP_MEMORY_SCOPE_AGENT);
+        u64 rw[3] = {0, 0, 0};
+        if (found != 0) {
+          for (int i = 0; i < 10; i++) {
+            rw[i >> 2] |= static_cast<u64>(dev->ctl.res[i]) << ((i & 3) * 16);
+          }
+        }
+        // Response: relaxed system stores (write-through), drained before
+        // the resp_seq flag store.
+        __hip_atomic_store(as_gu64(&mb->r_evaluated), ev, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+        __hip_atomic_store(as_gu32(&mb->r_found), found, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+        for (int i = 0; i < 3; i++) {
+          __hip_atomic_store(as_gu64(&mb->r_res_w[i]), rw[i], __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_SYSTEM);
+        }
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __hip_atomic_store(as_gu64(&mb->resp_seq), seq, __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_SYSTEM);
       }
     }
@@ -1498,21 +1523,13 @@ ScanResult ScanService::scan4(const ScanRequest& rq, i64 begin, i64 end) {
     const auto dt = std::chrono::steady_clock::now() - t0;
     if (svc_debug() && !reported && dt > std::chrono::seconds(1)) {
       reported = true;
-      int hist[8] = {};
-      for (int i = 0; i < grid_ && i < 4096; i++) {
-        hist[mb_->stage[i] & 7] += 1;
-      }
       std::fprintf(stderr,
                    "[svc] slow request: seq=%llu resp=%llu state=%u alive=%u "
-                   "n=%d range=%lld polls=%llu seen_rs=%llu stages "
-                   "[%d %d %d %d %d %d]\n",
+                   "n=%d range=%lld\n",
                    static_cast<unsigned long long>(s),
                    static_cast<unsigned long long>(mb_->resp_seq),
                    mb_->svc_state, mb_->alive, rq.n,
-                   static_cast<long long>(end - begin),
-                   static_cast<unsigned long long>(mb_->leader_polls),
-                   static_cast<unsigned long long>(mb_->leader_seen_rs),
-                   hist[0], hist[1], hist[2], hist[3], hist[4], hist[5]);
+                   static_cast<long long>(end - begin));
     }
     if (!escalated && dt > std::chrono::seconds(10)) {
       escalated = true;
@@ -1535,7 +1552,9 @@ ScanResult ScanService::scan4(const ScanRequest& rq, i64 begin, i64 end) {
   out.evaluated = mb_->r_evaluated;
   if (mb_->r_found != 0) {
     out.found = true;
-    std::memcpy(out.res, mb_->r_res, sizeof(out.res));
+    for (int i = 0; i < 10; i++) {
+      out.res[i] = static_cast<u16>(mb_->r_res_w[i >> 2] >> ((i & 3) * 16));
+    }
   }
   return out;
 }
