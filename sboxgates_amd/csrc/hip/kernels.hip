@@ -1031,11 +1031,21 @@ struct SvcDev {  // device memory
   alignas(64) Avail3Matcher matcher;
 };
 
-// ~2 ms of leader poll iterations (each ~1-2 us: one PCIe read + s_sleep).
-constexpr int SVC_IDLE_POLLS = 1500;
+// Leader poll iterations before idle retirement (each ~1-2 us: one PCIe
+// read + s_sleep) — about a second. Gate-mode searches interleave CPU
+// scans between service requests, so a short timeout causes relaunch
+// storms; in-process k=5/7 launches park the service explicitly instead.
+inline int svc_idle_polls() {
+  static const int v = [] {
+    const char* s = std::getenv("SBOXGATES_SVC_IDLE");
+    return s != nullptr ? std::atoi(s) : 500000;
+  }();
+  return v;
+}
 
 __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
-                                                              SvcDev* dev) {
+                                                              SvcDev* dev,
+                                                              int c_idle_polls) {
   __shared__ alignas(16) u64 s_pool[MAX_GATES * PSTR];
   __shared__ alignas(16) u8 s_bitmap[256 * 256 / 8];
   __shared__ u8 s_funs[256];
@@ -1073,7 +1083,7 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
             word = served + 3;  // bump seq, set quit bit
             break;
           }
-          if (++idle > SVC_IDLE_POLLS) {
+          if (++idle > c_idle_polls) {
             // Retire. After this store the kernel never touches the
             // mailbox again; the host reaps and relaunches on demand.
             __hip_atomic_store(as_gu32(&mb->svc_state), SVC_RETIRED,
@@ -1124,7 +1134,11 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
         if (epoch != dev->matcher_epoch) {
           const u64* ms = reinterpret_cast<const u64*>(&mb->matcher_staging);
           u64* md = reinterpret_cast<u64*>(&dev->matcher);
-          for (size_t i = threadIdx.x; i < sizeof(Avail3Matcher) / 8;
+          // Round UP: sizeof(Avail3Matcher) is not a multiple of 8 and the
+          // last partial word holds `count`; truncating it left s_count = 0
+          // and made every early-exit probe silently miss. Both structs sit
+          // in 64-aligned tails, so the overread is in-allocation padding.
+          for (size_t i = threadIdx.x; i < (sizeof(Avail3Matcher) + 7) / 8;
                i += blockDim.x) {
             md[i] = ms[i];
           }
@@ -1314,7 +1328,14 @@ ScanService::ScanService(int device) : device_(device) {
       &per_cu, reinterpret_cast<const void*>(k_scan4_service), SCAN_BLOCK, 0));
   hipDeviceProp_t prop;
   SBG_HIP_CHECK(hipGetDeviceProperties(&prop, device_));
-  grid_ = std::clamp(per_cu * prop.multiProcessorCount, 8, 2048);
+  // Small grid by design: the scans the service exists for are <= ~3e5
+  // candidates (deep-recursion step-4 scans), and the request/response
+  // round trip scales with the completion barrier — the flat-counter
+  // barrier price is ~7-26 us at 256-1024 workgroups
+  // (MI355X_MICROARCH.md barrier-counter) but single-digit at 128. A
+  // 128-WG grid also coexists with other kernels (half the CUs stay
+  // empty), so an idle-resident service cannot starve them.
+  grid_ = std::clamp(std::min(per_cu * prop.multiProcessorCount, 128), 8, 2048);
   if (const char* g = std::getenv("SBOXGATES_SVC_GRID")) {
     long v = std::strtol(g, nullptr, 10);
     if (v >= 1 && v <= 4096) grid_ = static_cast<int>(v);
@@ -1409,7 +1430,7 @@ void ScanService::ensure_running_locked() {
   mb_->alive = 0;
   __atomic_store_n(&mb_->svc_state, SVC_RUNNING, __ATOMIC_RELEASE);
   hipLaunchKernelGGL(k_scan4_service, dim3(grid_), dim3(SCAN_BLOCK), 0, stream_,
-                     mb_, d_svc_);
+                     mb_, d_svc_, svc_idle_polls());
   SBG_HIP_CHECK(hipGetLastError());
   running_ = true;
   relaunches_ += 1;
