@@ -19,7 +19,7 @@ PY_INC     := $(shell $(PYTHON) -c "import sysconfig; print(sysconfig.get_paths(
 PYBIND_INC := $(shell $(PYTHON) -c "import pybind11; print(pybind11.get_include())")
 EXT_SUFFIX := $(shell $(PYTHON) -c "import sysconfig; print(sysconfig.get_config_var('EXT_SUFFIX'))")
 
-CXXFLAGS   := -O3 -std=c++20 -fPIC -Wall -Wextra -Wno-unused-parameter -MMD -MP $(INC)
+CXXFLAGS   := -O3 -mavx2 -std=c++20 -fPIC -Wall -Wextra -Wno-unused-parameter -MMD -MP $(INC)
 HIPFLAGS   := --offload-arch=$(GPU_ARCH)
 LDFLAGS    := -shared -fPIC
 

@@ -275,16 +275,29 @@ ScanResult Engine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   // and tripled the wall time. k=4 scans go through the persistent scan
   // service (~5-8 us/call vs the ~30 us one-shot floor), which moves the
   // cutover well down.
-  static const i64 min4_svc = [] {
+  // k=4 via the scan service: adaptive cutover. The service round trip is
+  // roughly constant (~SVC_US), while the CPU path's per-candidate cost
+  // varies ~10x with mask sparsity; route to the GPU when the estimated
+  // CPU time for the range exceeds the round trip. SBOXGATES_GPU_MIN4
+  // pins a fixed threshold instead.
+  static const i64 min4_fixed = [] {
     const char* s = std::getenv("SBOXGATES_GPU_MIN4");
-    return s != nullptr ? std::strtoll(s, nullptr, 10) : 4096;
+    return s != nullptr ? std::strtoll(s, nullptr, 10) : -1;
   }();
+  i64 min4 = min4_fixed;
+  if (min4 < 0) {
+    constexpr double SVC_US = 12e-6;
+    const double rate = stats_.scan_seconds3_cpu > 1e-4
+                            ? stats_.candidates3_cpu / stats_.scan_seconds3_cpu
+                            : 2e8;
+    min4 = std::clamp<i64>(static_cast<i64>(rate * SVC_US), 1024, 1 << 16);
+  }
   const i64 gpu_min =
       k == 7 ? 256
              : (k == 5 ? 4096
                        : (k == 4 && gpu_ != nullptr &&
                                   gpu_->scan4_service_active()
-                              ? min4_svc
+                              ? min4
                               : 1 << 14));
   bool use_gpu = gpu_ != nullptr && (opt_.gpu == GPU_FORCE || end - begin >= gpu_min);
   const auto t0 = std::chrono::steady_clock::now();
@@ -304,6 +317,10 @@ ScanResult Engine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   }
   const double dt =
       std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
+  if (!use_gpu && k <= 4) {
+    stats_.candidates3_cpu += r.evaluated;
+    stats_.scan_seconds3_cpu += dt;
+  }
   switch (k) {
     case 3:
     case 4:

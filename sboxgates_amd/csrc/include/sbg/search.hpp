@@ -36,6 +36,11 @@ struct SearchStats {
   double scan_seconds3 = 0;  // wall time inside 3-input scans (incl. k=4)
   double scan_seconds5 = 0;
   double scan_seconds7 = 0;
+  // CPU-path-only k=3/4 totals: feed the adaptive GPU cutover (the CPU
+  // rate swings ~10x between full and sparse masks, so the size threshold
+  // where the scan-service round trip wins moves with it).
+  u64 candidates3_cpu = 0;
+  double scan_seconds3_cpu = 0;
 };
 
 class Engine {
