@@ -1,46 +1,48 @@
 cd /root/repo
 export HSA_ENABLE_IPC_MODE_LEGACY=0
 mkdir -p gpurun_out
-SB=sboxgates_amd/sboxes/rijndael.txt
-cat > /tmp/svc_micro.py <<'PY'
-import sys, time, os
+echo "=== trial-17 repro ==="
+timeout 120 python -u - <<'PY' 2>&1 | tail -12
+import sys, struct, random
 sys.path.insert(0, '.')
 from sboxgates_amd import models
 from sboxgates_amd.ops import make_engine, mask_for_inputs
-eng = make_engine(gpu="force", seed=1)
+gpu = make_engine(lut_graph=True, seed=1, gpu="force", save_states=False)
+cpu = make_engine(lut_graph=True, seed=1, gpu="off", save_states=False)
 sbox, n = models.load("rijndael")
-eng.set_sbox(sbox, n)
-st = eng.initial_state()
-st.grow_pool_random(80, 7)
-t = eng.target(0); mask = mask_for_inputs(8)
-mode = "nosvc" if os.environ.get("SBOXGATES_NO_SVC") else "svc"
-for rng in (512, 2048, 16384, 82160):
-    for _ in range(30): eng.scan_pool(4, st, t, mask, 0, rng, 1, True)
-    N = 2000
-    t0 = time.perf_counter()
-    for i in range(N):
-        f, r, ev = eng.scan_pool(4, st, t, mask, 0, rng, i, True)
-        assert ev == rng
-    dt = (time.perf_counter() - t0) / N
-    print(f"[{mode}] k4 range={rng}: {dt*1e6:.1f} us/call (python-inclusive)", flush=True)
-# pool-delta path: alternate pools to exercise prefix sync
-st2 = eng.initial_state(); st2.grow_pool_random(80, 8)
-t0 = time.perf_counter(); N = 1000
-for i in range(N):
-    f, r, ev = eng.scan_pool(4, st if i % 2 else st2, t, mask, 0, 2048, i, True)
-    assert ev == 2048
-print(f"[{mode}] k4 alternating pools range=2048: {(time.perf_counter()-t0)/N*1e6:.1f} us/call", flush=True)
+gpu.set_sbox(sbox, n); cpu.set_sbox(sbox, n)
+# reconstruct trial 17's exact inputs
+rng = random.Random(0xF022)
+for trial in range(18):
+    k = rng.choice([3, 4, 5, 5, 7, 7])
+    pool = rng.choice([15, 25, 40, 70])
+    if k == 7 and pool > 40: pool = 40
+    seedbits = rng.getrandbits(32)
+    kind = rng.randrange(3)
+    if kind == 0:
+        words = [2**64-1]*4
+        maskbits = None
+    else:
+        words = [0,0,0,0]
+        nbits = rng.choice([4, 12, 40, 150])
+        for _ in range(nbits):
+            i = rng.randrange(256); words[i//64] |= 1 << (i % 64)
+    tbit = rng.randrange(8)
+    import math
+    total = math.comb(pool, 3 if k == 4 else k)
+    a = rng.randrange(total)
+    b = min(total, a + rng.choice([3, 500, 30_000, 200_000]))
+st = gpu.initial_state(); st.grow_pool_random(pool, seedbits)
+mask = struct.pack("<4Q", *words)
+target = gpu.target(tbit)
+print("trial17 config:", k, pool, hex(seedbits), tbit, a, b)
+f_g, r_g, ev_g = gpu.scan_pool(k, st, target, mask, a, b, 17, True)
+f_c, r_c, ev_c = cpu.scan_pool(k, st, target, mask, a, b, 17, True)
+print("count_all: ev_g=%d ev_c=%d want=%d f_g=%s f_c=%s" % (ev_g, ev_c, b-a, f_g, f_c))
+f_g, r_g, ev_g = gpu.scan_pool(k, st, target, mask, a, b, 17)
+f_c, r_c, ev_c = cpu.scan_pool(k, st, target, mask, a, b, 17)
+print("early-exit: f_g=%s f_c=%s r_g=%s r_c=%s" % (f_g, f_c, r_g[:5], r_c[:5]))
 PY
-echo "=== micro svc ==="
-timeout 240 python -u /tmp/svc_micro.py 2>&1 | tail -6
-echo "=== micro nosvc ==="
-SBOXGATES_NO_SVC=1 timeout 240 python -u /tmp/svc_micro.py 2>&1 | tail -6
-echo "=== k4 parity tests ==="
-timeout 600 python -m pytest tests/test_gpu.py::test_scan4_gpu_matches_cpu tests/test_gpu.py::test_gpu_window_fuzz_soak tests/test_gpu.py::test_random_window_count_parity -x -q 2>&1 | tail -3
-echo "=== gate-mode AES bit 0 (auto hybrid) ==="
-timeout 300 bash -c "time ./bin/sboxgates -o 0 --seed 11 -v $SB" > gpurun_out/gate_auto.log 2>&1
-echo "rc=$?"; tail -5 gpurun_out/gate_auto.log
-echo "=== gate-mode AES bit 0 (auto, no service) ==="
-SBOXGATES_NO_SVC=1 timeout 300 bash -c "time ./bin/sboxgates -o 0 --seed 11 -v $SB" > gpurun_out/gate_auto_nosvc.log 2>&1
-echo "rc=$?"; tail -5 gpurun_out/gate_auto_nosvc.log
+echo "=== full GPU suite ==="
+timeout 1100 python -m pytest tests/test_gpu.py -q 2>&1 | tail -12
 echo ALL_DONE

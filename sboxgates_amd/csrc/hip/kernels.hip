@@ -1335,7 +1335,7 @@ ScanService::ScanService(int device) : device_(device) {
   // (MI355X_MICROARCH.md barrier-counter) but single-digit at 128. A
   // 128-WG grid also coexists with other kernels (half the CUs stay
   // empty), so an idle-resident service cannot starve them.
-  grid_ = std::clamp(std::min(per_cu * prop.multiProcessorCount, 128), 8, 2048);
+  grid_ = std::clamp(std::min(per_cu * prop.multiProcessorCount, 64), 8, 2048);
   if (const char* g = std::getenv("SBOXGATES_SVC_GRID")) {
     long v = std::strtol(g, nullptr, 10);
     if (v >= 1 && v <= 4096) grid_ = static_cast<int>(v);
@@ -1693,7 +1693,10 @@ ScanResult GpuEngine::scan(int k, const ScanRequest& rq, i64 begin, i64 end) {
   Impl* im = impl_;
   SBG_HIP_CHECK(hipSetDevice(im->device));
 
-  if (k == 4 && scan4_service_active()) {
+  // Service only for small/medium ranges: its 64-WG grid is sized for the
+  // request/response round trip, not for multi-million-candidate scans —
+  // those go to the one-shot kernel with a work-sized grid.
+  if (k == 4 && end - begin <= (1 << 20) && scan4_service_active()) {
     if (rq.matcher == nullptr) throw std::runtime_error("scan4 needs matcher");
     try {
       return im->svc->scan4(rq, begin, end);
