@@ -58,7 +58,9 @@ def main():
     if distributed:
         backend = "nccl" if use_gpu else "gloo"
         if use_gpu:
-            torch.cuda.set_device(local_rank)
+            # modulo: lets oversubscribed rehearsals (world > device count,
+            # all ranks on one device) run the same code path.
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group(backend=backend)
 
     if not use_gpu and not args.allow_cpu:

@@ -13,6 +13,7 @@
 // multi-process deployment.
 
 #include <getopt.h>
+#include <dirent.h>
 #include <sys/stat.h>
 #include <sys/types.h>
 
@@ -68,6 +69,9 @@ void print_help(const char* prog) {
       "      --cpu                     Disable the GPU path.\n"
       "      --gpu                     Require the GPU path (fail if no device).\n"
       "      --output-dir=DIR          Directory for XML checkpoint files.\n"
+      "      --resume-dir=DIR          Write checkpoints to DIR and resume the\n"
+      "                                search from the best state file already\n"
+      "                                in it (most outputs, then fewest gates).\n"
       "      --gpus=N                  Drive N GPUs from this process (one\n"
       "                                worker thread per device).\n"
       "      --beam=N                  Tied-state beam width for multi-output\n"
@@ -99,7 +103,7 @@ int main(int argc, char** argv) {
   opt.set_avail_gates(sbg::DEFAULT_GATE_BITFIELD);
 
   enum { OPT_HIP = 1000, OPT_SEED, OPT_CPU, OPT_GPU, OPT_OUTDIR, OPT_HELP,
-         OPT_GPUS, OPT_BEAM, OPT_JOBS, OPT_MAXG };
+         OPT_GPUS, OPT_BEAM, OPT_JOBS, OPT_MAXG, OPT_RESUME };
   static const struct option long_opts[] = {
       {"available-gates", required_argument, nullptr, 'a'},
       {"convert-c", no_argument, nullptr, 'c'},
@@ -117,6 +121,7 @@ int main(int argc, char** argv) {
       {"cpu", no_argument, nullptr, OPT_CPU},
       {"gpu", no_argument, nullptr, OPT_GPU},
       {"output-dir", required_argument, nullptr, OPT_OUTDIR},
+      {"resume-dir", required_argument, nullptr, OPT_RESUME},
       {"gpus", required_argument, nullptr, OPT_GPUS},
       {"beam", required_argument, nullptr, OPT_BEAM},
       {"jobs", required_argument, nullptr, OPT_JOBS},
@@ -129,6 +134,7 @@ int main(int argc, char** argv) {
   char* endptr = nullptr;
   long v;
   int max_gates_bound = -1;  // --max-gates: bound on TOTAL gates (incl. inputs)
+  std::string resume_dir;    // --resume-dir: pick up the best state in DIR
   while ((ch = getopt_long(argc, argv, "a:cdg:i:lno:p:svV", long_opts, nullptr)) != -1) {
     switch (ch) {
       case 'a':
@@ -173,6 +179,11 @@ int main(int argc, char** argv) {
       case OPT_OUTDIR:
         opt.output_dir = optarg;
         (void)mkdir(optarg, 0755);  // best-effort; open errors surface later
+        break;
+      case OPT_RESUME:
+        opt.output_dir = optarg;
+        resume_dir = optarg;
+        (void)mkdir(optarg, 0755);
         break;
       case OPT_GPUS:
         v = std::strtol(optarg, &endptr, 10);
@@ -297,7 +308,42 @@ int main(int argc, char** argv) {
     }
 
     sbg::state st;
-    if (opt.gfname.empty()) {
+    if (!resume_dir.empty() && opt.gfname.empty()) {
+      // Resume: scan DIR for state files and pick the most advanced one
+      // (most wired outputs, then fewest gates). The beam driver then
+      // continues adding outputs from that state — checkpoint/restart
+      // across budget windows for long multi-output runs.
+      engine.initial_state(st);
+      int best_outputs = -1;
+      std::string best_file;
+      if (DIR* d = opendir(resume_dir.c_str())) {
+        while (struct dirent* e = readdir(d)) {
+          std::string name = e->d_name;
+          if (name.size() < 5 || name.substr(name.size() - 4) != ".xml") continue;
+          sbg::state cand;
+          std::string lerr;
+          if (!sbg::load_state(resume_dir + "/" + name, &cand, &lerr)) continue;
+          if (sbg::get_num_inputs(&cand) != static_cast<int>(num_inputs)) continue;
+          int outs = 0;
+          for (int i = 0; i < 8; i++) {
+            if (cand.outputs[i] != sbg::NO_GATE) outs += 1;
+          }
+          if (outs > best_outputs ||
+              (outs == best_outputs && cand.num_gates < st.num_gates)) {
+            best_outputs = outs;
+            st = cand;
+            best_file = name;
+          }
+        }
+        closedir(d);
+      }
+      if (best_outputs > 0) {
+        std::printf("Resuming from %s/%s (%d output%s, %d gates).\n",
+                    resume_dir.c_str(), best_file.c_str(), best_outputs,
+                    best_outputs == 1 ? "" : "s",
+                    st.num_gates - sbg::get_num_inputs(&st));
+      }
+    } else if (opt.gfname.empty()) {
       engine.initial_state(st);
     } else if (!sbg::load_state(opt.gfname, &st, &err)) {
       return fail("Error when reading state file", err.c_str());
