@@ -1120,8 +1120,12 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
         // Pull the request: header block, pool delta, matcher delta.
         if (threadIdx.x < HDR_WORDS) s_hdr[threadIdx.x] = mb->hdr[threadIdx.x];
         __syncthreads();
-        const int n = static_cast<int>(s_hdr[HDR_N_KEEP] & 0xFFFFFFFFu);
-        const int keep = static_cast<int>(s_hdr[HDR_N_KEEP] >> 32);
+        // Defensive clamps: a corrupt header must never walk past the
+        // fixed-size pool buffers (a GPU page fault can wedge the node).
+        const int n = std::min(
+            static_cast<int>(s_hdr[HDR_N_KEEP] & 0xFFFFFFFFu), MAX_GATES);
+        const int keep = std::max(
+            0, std::min(static_cast<int>(s_hdr[HDR_N_KEEP] >> 32), n));
         const u32 epoch = static_cast<u32>(s_hdr[HDR_EPOCH_CALL] & 0xFFFFFFFFu);
         if (threadIdx.x < HDR_WORDS) dev->hdr[threadIdx.x] = s_hdr[threadIdx.x];
         {
@@ -1177,7 +1181,8 @@ __global__ void __launch_bounds__(SCAN_BLOCK) k_scan4_service(SvcMailbox* mb,
       if (threadIdx.x < HDR_WORDS) s_hdr[threadIdx.x] = dev->hdr[threadIdx.x];
     }
     __syncthreads();
-    const int n = static_cast<int>(s_hdr[HDR_N_KEEP] & 0xFFFFFFFFu);
+    const int n = std::min(static_cast<int>(s_hdr[HDR_N_KEEP] & 0xFFFFFFFFu),
+                           MAX_GATES);
     const int count_all = static_cast<int>(s_hdr[HDR_EPOCH_CALL] >> 32);
     const i64 begin = static_cast<i64>(s_hdr[HDR_BEGIN]);
     const i64 end = static_cast<i64>(s_hdr[HDR_END]);
