@@ -172,3 +172,23 @@ def test_max_gates_bound(tmp_path):
     assert r.returncode == 0
     assert glob.glob(os.path.join(str(tmp_path), "1-*.xml"))
     assert run(["--max-gates", "0", des()]).returncode != 0
+
+
+def test_resume_dir(tmp_path):
+    """--resume-dir: a second invocation picks up the most advanced state
+    from the first (checkpoint/restart across budget windows)."""
+    d = os.path.join(str(tmp_path), "ckpt")
+    r = run(["-a", "10694", "-p", "63", "--seed", "5", "--cpu",
+             "--resume-dir", d, des()], cwd=str(tmp_path), timeout=300)
+    assert r.returncode == 0, r.stderr
+    first = set(glob.glob(os.path.join(d, "*.xml")))
+    assert first, "first run produced no checkpoints"
+    r2 = run(["-a", "10694", "-p", "63", "--seed", "6", "--cpu",
+              "--resume-dir", d, des()], cwd=str(tmp_path), timeout=300)
+    assert r2.returncode == 0, r2.stderr
+    assert "Resuming from" in r2.stdout
+    # It resumed from a full 4-output state, so the second run starts (and
+    # ends) complete without redoing outputs.
+    import re
+    m = re.search(r"Resuming from \S+ \((\d+) outputs?", r2.stdout)
+    assert m and int(m.group(1)) >= 1
