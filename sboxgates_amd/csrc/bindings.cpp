@@ -441,6 +441,10 @@ PYBIND11_MODULE(_core, m) {
         d["scan_seconds3"] = e.stats().scan_seconds3;
         d["scan_seconds5"] = e.stats().scan_seconds5;
         d["scan_seconds7"] = e.stats().scan_seconds7;
+        d["nodes"] = e.stats().nodes;
+        d["step12_seconds"] = e.stats().step12_seconds;
+        d["step3_seconds"] = e.stats().step3_seconds;
+        d["step4a_seconds"] = e.stats().step4a_seconds;
         return d;
       })
       .def("scan_pool", [](Engine& e, int k, const state& st, py::bytes target,

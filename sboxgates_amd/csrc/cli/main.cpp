@@ -375,6 +375,10 @@ int main(int argc, char** argv) {
                   rate(st_.candidates7, st_.scan_seconds7),
                   (unsigned long long)st_.gpu_scans,
                   (unsigned long long)st_.cpu_scans);
+      std::printf("Host phases: %llu nodes; step1/2 %.2fs, step3 %.2fs, "
+                  "step4a %.2fs\n",
+                  (unsigned long long)st_.nodes, st_.step12_seconds,
+                  st_.step3_seconds, st_.step4a_seconds);
     }
   } catch (const std::exception& e) {
     std::fprintf(stderr, "%s\n", e.what());

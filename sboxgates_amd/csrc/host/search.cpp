@@ -556,6 +556,8 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
 
   const ttable T1 = target & mask;
   const ttable T0 = ~target & mask;
+  stats_.nodes += 1;
+  const auto t_s12 = std::chrono::steady_clock::now();
 
   // Step 1: an existing gate already realizes the map (sboxgates.c:301-308).
   for (int i = 0; i < st->num_gates; i++) {
@@ -574,6 +576,9 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
                         mask, "step2");
     }
   }
+  stats_.step12_seconds +=
+      std::chrono::duration<double>(std::chrono::steady_clock::now() - t_s12)
+          .count();
 
   // Step 3: a pair combined by one available gate (sboxgates.c:323-350).
   // Implemented via 4-cell forced-bit requirements instead of per-function
@@ -581,6 +586,17 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
   if (!check_num_gates_possible(st, 1, sat_metric_of(AND), opt_.metric)) {
     return NO_GATE;
   }
+  struct PhaseTimer {
+    double* acc;
+    std::chrono::steady_clock::time_point t0;
+    explicit PhaseTimer(double* a) : acc(a), t0(std::chrono::steady_clock::now()) {}
+    ~PhaseTimer() {
+      *acc += std::chrono::duration<double>(std::chrono::steady_clock::now() - t0)
+                  .count();
+    }
+  };
+  {
+  PhaseTimer pt3(&stats_.step3_seconds);
   for (int i = 0; i < st->num_gates; i++) {
     const gatenum gi = gate_order[i];
     const ttable& ti = st->gates[gi].table;
@@ -604,6 +620,7 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
       }
     }
   }
+  }
 
   if (opt_.lut_graph) {
     gatenum ret = lut_search(st, target, mask, inbits, gate_order);
@@ -614,6 +631,8 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
                                   opt_.metric)) {
       return NO_GATE;
     }
+    {
+    PhaseTimer pt4a(&stats_.step4a_seconds);
     for (int i = 0; i < st->num_gates; i++) {
       const gatenum gi = gate_order[i];
       const ttable& ti = st->gates[gi].table;
@@ -638,6 +657,7 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
           }
         }
       }
+    }
     }
 
     // Step 4b: triples realized by an available composed 3-input function

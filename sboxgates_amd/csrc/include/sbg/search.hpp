@@ -41,6 +41,12 @@ struct SearchStats {
   // where the scan-service round trip wins moves with it).
   u64 candidates3_cpu = 0;
   double scan_seconds3_cpu = 0;
+  // Host-side phase timers for the gate-mode recursion (where the wall
+  // that is NOT scan time goes; see profiles/gate_mode_service.md).
+  u64 nodes = 0;              // create_circuit invocations
+  double step12_seconds = 0;  // step 1/2 existing-gate and inverse sweeps
+  double step3_seconds = 0;   // step 3 pair loop (host)
+  double step4a_seconds = 0;  // step 4a NOT-augmented pair loop (host)
 };
 
 class Engine {
