@@ -182,16 +182,36 @@ static gatenum assert_ret(gatenum ret, const ttable& target, const state* st,
 // a candidate pair/triple imposes, and permutations thereof.
 // ---------------------------------------------------------------------------
 
+// Per-gate masked halves, precomputed once per node for the step-3/4a
+// pair loops: p = t&T1, q = ~t&T1, r = t&T0, s = ~t&T0. A pair cell's
+// masked content is then one AND of two halves ((±tx)&(±ty)&T1 =
+// (±tx&T1)&(±ty&T1)), which removes the per-pair cell construction and
+// the ~t materializations — the pair loops are ~60% of gate-mode host
+// time (profiles/gate_mode_service.md).
+struct GateHalves {
+  ttable p, q, r, s;
+};
+
+static void fill_halves(GateHalves* out, const state* st, int n, const ttable& T1,
+                        const ttable& T0) {
+  for (int i = 0; i < n; i++) {
+    const ttable& t = st->gates[i].table;
+    out[i].p = t & T1;
+    out[i].q = ~t & T1;
+    out[i].r = t & T0;
+    out[i].s = ~t & T0;
+  }
+}
+
 // 4-cell requirements for a pair (x, y): bit p of req*/care is pattern
 // p = vx<<1 | vy. Returns false if some cell is contradictory (no 2-input
 // function can match).
-static bool pair_requirements(const ttable& tx, const ttable& ty, const ttable& T1,
-                              const ttable& T0, u8* req1, u8* care) {
+static bool pair_requirements(const GateHalves& x, const GateHalves& y, u8* req1,
+                              u8* care) {
   u8 r1 = 0, r0 = 0;
   for (int p = 0; p < 4; p++) {
-    ttable cell = (p & 2 ? tx : ~tx) & (p & 1 ? ty : ~ty);
-    bool has1 = tt_any(cell & T1);
-    bool has0 = tt_any(cell & T0);
+    bool has1 = tt_any((p & 2 ? x.p : x.q) & (p & 1 ? y.p : y.q));
+    bool has0 = tt_any((p & 2 ? x.r : x.s) & (p & 1 ? y.r : y.s));
     if (has1 && has0) return false;
     if (has1) r1 |= 1u << p;
     if (has0) r0 |= 1u << p;
@@ -595,16 +615,17 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
                   .count();
     }
   };
+  static thread_local std::vector<GateHalves> halves;
+  halves.resize(st->num_gates);
+  fill_halves(halves.data(), st, st->num_gates, T1, T0);
   {
   PhaseTimer pt3(&stats_.step3_seconds);
   for (int i = 0; i < st->num_gates; i++) {
     const gatenum gi = gate_order[i];
-    const ttable& ti = st->gates[gi].table;
     for (int k = i + 1; k < st->num_gates; k++) {
       const gatenum gk = gate_order[k];
-      const ttable& tk = st->gates[gk].table;
       u8 req1, care;
-      if (!pair_requirements(ti, tk, T1, T0, &req1, &care)) continue;
+      if (!pair_requirements(halves[gi], halves[gk], &req1, &care)) continue;
       const u8 req1s = swap_pair_patterns(req1);
       const u8 cares = swap_pair_patterns(care);
       for (int m = 0; opt_.avail_gates[m].num_inputs != 0; m++) {
@@ -631,16 +652,18 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
                                   opt_.metric)) {
       return NO_GATE;
     }
-    {
+    // Without -n the NOT-augmented function list is empty: the whole pair
+    // sweep would compute requirements for zero candidate functions (the
+    // reference pays this too, sboxgates.c:366-386 — measured at 22 s of
+    // the AES bit-0 CPU run).
+    if (opt_.avail_not[0].num_inputs != 0) {
     PhaseTimer pt4a(&stats_.step4a_seconds);
     for (int i = 0; i < st->num_gates; i++) {
       const gatenum gi = gate_order[i];
-      const ttable& ti = st->gates[gi].table;
       for (int k = i + 1; k < st->num_gates; k++) {
         const gatenum gk = gate_order[k];
-        const ttable& tk = st->gates[gk].table;
         u8 req1, care;
-        if (!pair_requirements(ti, tk, T1, T0, &req1, &care)) continue;
+        if (!pair_requirements(halves[gi], halves[gk], &req1, &care)) continue;
         const u8 req1s = swap_pair_patterns(req1);
         const u8 cares = swap_pair_patterns(care);
         for (int m = 0; opt_.avail_not[m].num_inputs != 0; m++) {
