@@ -179,6 +179,9 @@ def main():
         print(json.dumps(out))
 
     if distributed:
+        # Ranks != 0 arrive here while rank 0 runs the gate search; leave
+        # together so no rank tears its communicator down under a peer.
+        dist.barrier()
         dist.destroy_process_group()
 
 
