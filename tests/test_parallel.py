@@ -72,3 +72,13 @@ def test_distributed_world3_small_chunks(tmp_path):
                  extra_env={"SBOXGATES_CHUNK5": "1500",
                             "SBOXGATES_CHUNK7": "15000"})
     assert all(res[r]["ok"] for r in range(3))
+
+
+def test_distributed_world4_small_chunks(tmp_path):
+    """World-4 multi-process search (the driver's 4-GPU shape on the gloo
+    transport), with chunk sizes forcing many allreduce rounds."""
+    res = launch("one_output_search", tmp_path, world=4,
+                 extra_env={"SBOXGATES_CHUNK5": "5000",
+                            "SBOXGATES_CHUNK7": "40000"})
+    assert res[0]["ok"]
+    assert res[0]["gates"] > 0
