@@ -1576,6 +1576,12 @@ ScanResult ScanService::scan4(const ScanRequest& rq, i64 begin, i64 end) {
 
   ScanResult out;
   out.evaluated = mb_->r_evaluated;
+  if (out.evaluated > static_cast<u64>(end - begin)) {
+    // A response that claims more work than the range is corruption;
+    // treat it like a service failure (caller falls back to one-shot).
+    quit_locked();
+    throw std::runtime_error("scan service returned corrupt counts");
+  }
   if (mb_->r_found != 0) {
     out.found = true;
     for (int i = 0; i < 10; i++) {

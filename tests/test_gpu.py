@@ -424,3 +424,23 @@ def test_parallel_jobs_on_gpu():
         assert r.returncode == 0, r.stderr
         import glob as g
         assert g.glob(os.path.join(d, "1-*.xml"))
+
+
+def test_concurrent_jobs_gate_mode_gpu():
+    """Gate-mode --jobs: several engines sharing the one persistent scan
+    service on one device (the config-2 full-graph deployment shape)."""
+    import subprocess
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cli = os.path.join(repo, "bin", "sboxgates")
+    if not os.path.exists(cli):
+        pytest.skip("CLI not built")
+    import tempfile
+    with tempfile.TemporaryDirectory() as d:
+        r = subprocess.run(
+            [cli, "-o", "0", "-i", "8", "--jobs", "4", "--gpu", "--seed", "7",
+             os.path.join(repo, "sboxgates_amd", "sboxes", "des_s1.txt")],
+            cwd=d, capture_output=True, text=True, timeout=420)
+        assert r.returncode == 0, r.stderr
+        import glob as g
+        assert g.glob(os.path.join(d, "1-*.xml"))
