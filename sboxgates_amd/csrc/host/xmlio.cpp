@@ -262,15 +262,26 @@ std::string state_to_xml(const state& st) {
 std::string save_state(const state& st, const std::string& dir) {
   std::string path = state_file_name(st);
   if (!dir.empty()) path = dir + "/" + path;
-  FILE* fp = std::fopen(path.c_str(), "w");
+  // Atomic checkpoint: write a temp file and rename into place, so a
+  // process killed mid-write (budget windows SIGKILL at arbitrary
+  // points) can never leave a truncated state file for --resume-dir to
+  // trip over. Found by the interrupt/resume soak (tools/resume_soak.py).
+  std::string tmp = path + ".tmp";
+  FILE* fp = std::fopen(tmp.c_str(), "w");
   if (fp == nullptr) {
-    std::fprintf(stderr, "sboxgates: error opening %s for writing\n", path.c_str());
+    std::fprintf(stderr, "sboxgates: error opening %s for writing\n", tmp.c_str());
     return "";
   }
   std::string xml = state_to_xml(st);
   bool ok = std::fwrite(xml.data(), 1, xml.size(), fp) == xml.size();
-  std::fclose(fp);
-  return ok ? path : "";
+  ok = std::fclose(fp) == 0 && ok;
+  if (ok) ok = std::rename(tmp.c_str(), path.c_str()) == 0;
+  if (!ok) {
+    std::remove(tmp.c_str());
+    std::fprintf(stderr, "sboxgates: error writing %s\n", path.c_str());
+    return "";
+  }
+  return path;
 }
 
 bool state_from_xml(const std::string& xml, state* out, std::string* err) {

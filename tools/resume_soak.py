@@ -77,7 +77,13 @@ def main():
         for name in os.listdir(ck):
             if not name.endswith(".xml"):
                 continue
-            st = _core.State.load(os.path.join(ck, name))
+            try:
+                st = _core.State.load(os.path.join(ck, name))
+            except RuntimeError:
+                # save_state() writes atomically (temp + rename), so a
+                # parse failure here is a bug, not an interruption scar.
+                raise AssertionError(
+                    f"trial {trial}: truncated/corrupt checkpoint {name}")
             outs = [b for b in range(8) if st.outputs[b] >= 0]
             if len(outs) > best_outs:
                 best, best_outs = st, len(outs)
