@@ -56,6 +56,13 @@ def main():
         ck = os.path.join(d, "ck")
         cmd = [CLI, "--cpu", "-a", vocab, "--beam", str(rng.choice([1, 2])),
                "--seed", str(rng.getrandbits(30)), "--resume-dir", ck, sfile]
+        mode = rng.randrange(3)
+        if mode == 1:
+            cmd.insert(1, "-l")          # LUT-graph mode
+        elif mode == 2 and vocab != "65535":
+            cmd.insert(1, "-s")          # SAT metric
+        if rng.randrange(4) == 0:
+            cmd[1:1] = ["-n"]            # NOT-augmented step 4a
         # Run with random kill windows until a run completes on its own.
         done = False
         for attempt in range(40):
