@@ -745,14 +745,38 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
     state nst;
     gatenum nst_out = NO_GATE;
 
+    // Cross-bit branch-and-bound (gate mode): once an earlier bit
+    // produced `best`, later bits' candidates are only ever KEPT when
+    // strictly smaller (ties keep the first), so bound their sub-searches
+    // to strictly fewer gates. Measured: 3.4x fewer recursion nodes /
+    // 3.4x wall on AES bit 0 (186 s -> 55 s CPU-only), at a ~0.4-gate
+    // mean cost on single-shot runs (the tight bound reorders the greedy
+    // exploration) that -i 2 at the higher speed more than recovers.
+    // Not applied in LUT mode (its searches are scan-dominated; quality
+    // of the headline LUT artifacts stays byte-for-byte reproducible)
+    // nor under the SAT metric (its bound check is pre-append, so a
+    // tight bound would not be strict). SBOXGATES_NO_PRUNE=1 disables.
+    static const bool prune_off = [] {
+      const char* e = std::getenv("SBOXGATES_NO_PRUNE");
+      return e != nullptr && e[0] != '\0' && e[0] != '0';
+    }();
+    const bool have_best = best.num_gates != 0;
+    const gatenum best_bound =
+        !prune_off && have_best && opt_.metric == METRIC_GATES &&
+                !opt_.lut_graph && best.num_gates >= 2
+            ? static_cast<gatenum>(best.num_gates - 2)
+            : MAX_GATES;
+
     if (opt_.lut_graph) {  // LUT-based multiplexer.
       copy_state(nst, *st);
+      if (nst.max_gates > best_bound) nst.max_gates = best_bound;
       nst.max_gates -= 1;  // Room for the multiplexer.
       gatenum fb = create_circuit(&nst, target, mask & ~fsel, next_inbits);
       if (fb == NO_GATE) continue;
       gatenum fc = create_circuit(&nst, target, mask & fsel, next_inbits);
       if (fc == NO_GATE) continue;
-      nst.max_gates += 1;
+      nst.max_gates = st->max_gates;  // restore the caller's bound exactly
+                                      // (it was clamped by best_bound)
 
       if (fb == fc) {
         nst_out = fb;
@@ -772,6 +796,7 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
     } else {  // Try both AND- and OR-based multiplexers; keep the smaller.
       state nst_and;
       copy_state(nst_and, *st);
+      if (nst_and.max_gates > best_bound) nst_and.max_gates = best_bound;
       nst_and.max_gates -= 2;
       nst_and.max_sat_metric -= sat_metric_of(AND) + sat_metric_of(XOR);
 
@@ -780,7 +805,7 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
       if (fb != NO_GATE) {
         gatenum fc = create_circuit(&nst_and, nst_and.gates[fb].table ^ target,
                                     mask & fsel, next_inbits);
-        nst_and.max_gates += 2;
+        nst_and.max_gates = st->max_gates;  // caller's bound (was clamped)
         nst_and.max_sat_metric += sat_metric_of(AND) + sat_metric_of(XOR);
         gatenum andg = add_and_gate(&nst_and, fc, static_cast<gatenum>(bit), opt_.metric);
         mux_out_and = add_xor_gate(&nst_and, fb, andg, opt_.metric);
@@ -788,6 +813,7 @@ gatenum Engine::create_circuit(state* st, const ttable& target, const ttable& ma
 
       state nst_or;
       copy_state(nst_or, *st);
+      if (nst_or.max_gates > best_bound) nst_or.max_gates = best_bound;
       if (mux_out_and != NO_GATE) {
         nst_or.max_gates = nst_and.num_gates;
         nst_or.max_sat_metric = nst_and.sat_metric;
