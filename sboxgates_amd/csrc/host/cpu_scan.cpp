@@ -12,6 +12,54 @@
 
 namespace sbg {
 
+// Pair-prefix cells for the k=3-shaped scans: lexicographic combination
+// order increments the third gate fastest, so the four (a,b) pair cells
+// (intersected with T1/T0, with liveness flags) are computed once per
+// prefix and each c costs only the final split — the same prefix-sharing
+// the gfx950 kernels use, on the CPU path.
+struct PairCells {
+  ttable H1[4], H0[4];
+  u8 nz1 = 0, nz0 = 0;
+
+  void fill(const ttable& ta, const ttable& tb, const ttable& T1,
+            const ttable& T0) {
+    nz1 = nz0 = 0;
+    for (int u = 0; u < 4; u++) {
+      ttable pc = (u & 2 ? ta : ~ta) & (u & 1 ? tb : ~tb);
+      H1[u] = pc & T1;
+      H0[u] = pc & T0;
+      if (tt_any(H1[u])) nz1 |= static_cast<u8>(1u << u);
+      if (tt_any(H0[u])) nz0 |= static_cast<u8>(1u << u);
+    }
+  }
+
+  // Equivalent of lut3_p_masks(ta, tb, tc, ...) under this prefix.
+  bool p_masks(const ttable& tc, u32* p1_out, u32* p0_out) const {
+    u32 p1 = 0, p0 = 0;
+    for (int u = 0; u < 4; u++) {
+      const bool l1 = (nz1 >> u) & 1, l0 = (nz0 >> u) & 1;
+      if (!l1 && !l0) continue;
+      u32 c1 = 0, c0 = 0;  // cell bits (u<<1)|vc with content
+      if (l1) {
+        ttable x = H1[u] & tc;
+        if (tt_any(x)) c1 |= 2;
+        if (tt_any(H1[u] ^ x)) c1 |= 1;
+      }
+      if (l0) {
+        ttable x = H0[u] & tc;
+        if (tt_any(x)) c0 |= 2;
+        if (tt_any(H0[u] ^ x)) c0 |= 1;
+      }
+      if (c1 & c0) return false;  // a cell forced both ways
+      p1 |= c1 << (u << 1);
+      p0 |= c0 << (u << 1);
+    }
+    *p1_out = p1;
+    *p0_out = p0;
+    return true;
+  }
+};
+
 static inline bool excluded(const ScanRequest& rq, const gatenum* nums, int k) {
   if (rq.excl_low64 == 0) return false;
   for (int i = 0; i < k; i++) {
@@ -31,13 +79,19 @@ ScanResult cpu_scan3(const ScanRequest& rq, i64 begin, i64 end) {
 
   gatenum nums[3];
   nth_combination(begin, rq.n, 3, 0, nums);
+  PairCells pc;
+  int pa = -1, pb = -1;
   for (i64 i = begin; i < end; i++) {
     out.evaluated++;
     // NOTE: the reference's 3-LUT scan does not reject inbits combinations
     // (lut.c:501-523) — parity kept: no exclusion here.
+    if (nums[0] != pa || nums[1] != pb) {
+      pa = nums[0];
+      pb = nums[1];
+      pc.fill(rq.tables[pa], rq.tables[pb], T1, T0);
+    }
     u32 p1, p0;
-    if (lut3_p_masks(rq.tables[nums[0]], rq.tables[nums[1]], rq.tables[nums[2]],
-                     T1, T0, &p1, &p0)) {
+    if (pc.p_masks(rq.tables[nums[2]], &p1, &p0)) {
       u8 func = lut3_function_from_p(p1, p0, hash_mix64(rq.seed ^ static_cast<u64>(i)));
       if (func != 0 && !rq.count_all) {
         out.found = true;
@@ -183,11 +237,17 @@ ScanResult cpu_scan4(const ScanRequest& rq, i64 begin, i64 end) {
 
   gatenum nums[3];
   nth_combination(begin, rq.n, 3, 0, nums);
+  PairCells pc;
+  int pa = -1, pb = -1;
   for (i64 i = begin; i < end; i++) {
     out.evaluated++;
+    if (nums[0] != pa || nums[1] != pb) {
+      pa = nums[0];
+      pb = nums[1];
+      pc.fill(rq.tables[pa], rq.tables[pb], T1, T0);
+    }
     u32 p1, p0;
-    if (lut3_p_masks(rq.tables[nums[0]], rq.tables[nums[1]], rq.tables[nums[2]],
-                     T1, T0, &p1, &p0)) {
+    if (pc.p_masks(rq.tables[nums[2]], &p1, &p0)) {
       const u8 req1 = static_cast<u8>(p1);
       const u8 care = static_cast<u8>(p1 | p0);
       for (int perm = 0; perm < 6 && !rq.count_all; perm++) {
