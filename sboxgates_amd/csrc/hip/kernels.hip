@@ -1022,7 +1022,8 @@ struct SvcMailbox {  // pinned fine-grained host memory
 
 struct SvcDev {  // device memory
   u64 dev_seq;   // published seq for non-leader workgroups
-  u32 dev_cmd;   // SVC_CMD_*
+  u32 dev_cmd;   // vestigial (quit now travels in dev_seq's low bit);
+                 // kept so SVC_DEV_CTL_BYTES spans a stable layout
   u32 matcher_epoch;
   unsigned long long done;
   DevCtl ctl;
